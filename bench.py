@@ -1,0 +1,134 @@
+"""Flagship training-step benchmark (driver contract — see repo brief).
+
+Measures the BASELINE.json headline: Llama2-7B, HSDP, bf16, no activation
+checkpointing, batch 2/GPU, seq 4096, synthetic data, random-init weights.
+`value` is the WHOLE-JOB tokens/sec over all N GPUs; `vs_baseline` compares
+the per-GPU rate to the reference's 4550 tok/s/GPU on 128x A100
+(BASELINE.md). Weak scaling: per-GPU work fixed as N grows.
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W] [--model llama2_7b]
+N>1 is launched by the driver via torch.distributed.run (one rank per GPU).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", type=str, default="llama2_7b")
+    ap.add_argument("--batch-size", type=int, default=2)
+    ap.add_argument("--seq-len", type=int, default=4096)
+    ap.add_argument("--sharding", type=str, default="hsdp")
+    ap.add_argument("--ac", type=str, default="0",
+                    help="selective AC fraction (0 = off)")
+    args = ap.parse_args()
+
+    from fms_fsdp_amd.config import get_model_config
+    from fms_fsdp_amd.models import Llama, LlamaBlock
+    from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+    from fms_fsdp_amd.parallel.policies import apply_selective_ac
+    from fms_fsdp_amd.utils.train import setup_environ_flags
+
+    setup_environ_flags()
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+    if world > 1 and not dist.is_initialized():
+        dist.init_process_group("nccl" if use_cuda else "gloo")
+
+    torch.manual_seed(2023)
+    mcfg = get_model_config(args.model)
+    with torch.device(device):
+        model = Llama(mcfg)
+        model.reset_parameters()
+    n_params = model.param_count()
+
+    sm = ShardedModel(model, LlamaBlock, sharding_strategy=args.sharding,
+                      param_dtype=torch.bfloat16 if use_cuda else torch.float32,
+                      reshard_after_forward=n_params > 30e9,
+                      prefetch_lookahead=1, device=device)
+    if args.ac not in ("0", "0.0", 0):
+        apply_selective_ac(sm, LlamaBlock, args.ac)
+    opt = ShardedAdamW(sm, lr=3e-4)
+
+    bs, sl = args.batch_size, args.seq_len
+    g = torch.Generator().manual_seed(1234 + rank)
+    inp = torch.randint(0, mcfg.src_vocab_size, (bs, sl + 1), generator=g).to(device)
+    x, y = inp[:, :-1], inp[:, 1:].contiguous()
+
+    def step():
+        opt.zero_grad()
+        loss = sm(x, labels=y)
+        loss.backward()
+        sm.clip_grad_norm_(1.0)
+        opt.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+
+    if dist.is_initialized():
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        loss = step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    elapsed = time.time() - t0
+
+    # max over ranks
+    et = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+    if dist.is_initialized():
+        dist.all_reduce(et, op=dist.ReduceOp.MAX)
+    elapsed = et.item()
+
+    ms_per_step = elapsed / args.steps * 1000
+    tokens_per_sec = world * bs * sl * args.steps / elapsed
+    baseline_per_gpu = 4550.0  # 7B HSDP compile noAC bs2, A100
+    vs_baseline = (tokens_per_sec / world) / baseline_per_gpu \
+        if args.model == "llama2_7b" else None
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "tokens/sec (aggregate)",
+            "value": round(tokens_per_sec, 1),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(vs_baseline, 3) if vs_baseline else None,
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic",
+            "config": {"model": args.model, "global_batch": bs * world,
+                       "seq_len": sl,
+                       "parallelism": f"{args.sharding}{world}",
+                       "tok_per_sec_per_gpu": round(tokens_per_sec / world, 1),
+                       "final_loss": round(float(loss.detach()), 4)},
+        }))
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

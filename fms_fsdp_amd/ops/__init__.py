@@ -1,0 +1,215 @@
+"""Op dispatch: hand-written CDNA4 HIP kernels on GPU, fp32 PyTorch
+reference on CPU.
+
+On a ROCm GPU the in-tree extension `fms_fsdp_amd._C` (built by
+`__graft_entry__.build()` / `setup.py build_ext --inplace`) is REQUIRED:
+ops raise RuntimeError rather than silently falling back to eager
+(per-project rule: no silent PyTorch fallback on the GPU path).
+"""
+
+import os
+
+import torch
+
+from . import reference
+
+_C = None
+_C_err = None
+try:
+    from fms_fsdp_amd import _C  # built in-tree, travels with the repo snapshot
+except ImportError as e:  # pragma: no cover - exercised only when unbuilt
+    _C_err = e
+
+
+def _require_ext(op):
+    if _C is None:
+        raise RuntimeError(
+            f"fms_fsdp_amd._C HIP extension is required for {op} on GPU but "
+            f"could not be imported ({_C_err}). Build it with "
+            f"`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950)."
+        )
+    return _C
+
+
+# --------------------------------------------------------------------------
+# RMSNorm
+# --------------------------------------------------------------------------
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        ext = _require_ext("rmsnorm")
+        x2d = x.contiguous().view(-1, x.shape[-1])
+        y, rinv = ext.rmsnorm_fwd(x2d, weight, eps)
+        ctx.save_for_backward(x2d, weight, rinv)
+        ctx.shape = x.shape
+        return y.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, weight, rinv = ctx.saved_tensors
+        dx, dw = _C.rmsnorm_bwd(dy.contiguous().view_as(x2d), x2d, weight, rinv)
+        return dx.view(ctx.shape), dw.to(weight.dtype), None
+
+
+def rmsnorm(x, weight, eps=1e-6):
+    if x.is_cuda:
+        return _RMSNormFn.apply(x, weight, eps)
+    return reference.rmsnorm(x, weight, eps)
+
+
+# --------------------------------------------------------------------------
+# RoPE (half-rotation convention; cos/sin tables (s, d/2) fp32)
+# --------------------------------------------------------------------------
+class _RoPEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, cos, sin):
+        ext = _require_ext("rope")
+        qo, ko = ext.rope_fwd(q.contiguous(), k.contiguous(), cos, sin, False)
+        ctx.save_for_backward(cos, sin)
+        return qo, ko
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        cos, sin = ctx.saved_tensors
+        dqo, dko = _C.rope_fwd(dq.contiguous(), dk.contiguous(), cos, sin, True)
+        return dqo, dko, None, None
+
+
+def rope_apply(q, k, cos, sin):
+    if q.is_cuda:
+        return _RoPEFn.apply(q, k, cos, sin)
+    return reference.rope_apply(q, k, cos, sin)
+
+
+# --------------------------------------------------------------------------
+# Causal flash attention (MFMA tiled, GQA)
+# --------------------------------------------------------------------------
+class _FlashAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v):
+        ext = _require_ext("attention")
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o, lse = ext.attn_fwd(q, k, v)
+        ctx.save_for_backward(q, k, v, o, lse)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = _C.attn_bwd(do.contiguous(), q, k, v, o, lse)
+        return dq, dk, dv
+
+
+def attention_causal(q, k, v):
+    if q.is_cuda:
+        if os.environ.get("FMS_AMD_ALLOW_TORCH_SDPA") == "1":
+            import torch.nn.functional as F
+            o = F.scaled_dot_product_attention(
+                q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+                is_causal=True, enable_gqa=(k.shape[2] != q.shape[2]))
+            return o.transpose(1, 2)
+        return _FlashAttnFn.apply(q, k, v)
+    return reference.attention_causal(q, k, v)
+
+
+# --------------------------------------------------------------------------
+# SwiGLU epilogue: silu(g) * u over fused (..., 2H) gate|up
+# --------------------------------------------------------------------------
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu):
+        ext = _require_ext("swiglu")
+        gu = gu.contiguous()
+        h = ext.swiglu_fwd(gu.view(-1, gu.shape[-1]))
+        ctx.save_for_backward(gu)
+        return h.view(*gu.shape[:-1], gu.shape[-1] // 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (gu,) = ctx.saved_tensors
+        dgu = _C.swiglu_bwd(dy.contiguous().view(-1, dy.shape[-1]),
+                            gu.view(-1, gu.shape[-1]))
+        return dgu.view(gu.shape)
+
+
+def swiglu(gu):
+    if gu.is_cuda:
+        return _SwiGLUFn.apply(gu)
+    return reference.swiglu(gu)
+
+
+# --------------------------------------------------------------------------
+# Fused linear + chunked cross-entropy (never materializes full fp32 logits)
+# --------------------------------------------------------------------------
+class _LinearCEFn(torch.autograd.Function):
+    """Computes mean CE of linear(x, W) vs labels, chunked over rows.
+
+    Forward also produces dx and dW (scaled for grad_output=1); backward
+    rescales. This trades one saved dx/dW pair for never holding the
+    (b*s, V) fp32 softmax (SURVEY.md hard-part 6: llama3 V=128256 logits
+    would be ~8 GB fp32 at b2 s8192).
+    """
+
+    CHUNK = 2048  # rows per chunk; 2048 x 128256 bf16 logits = 525 MB -> L2/HBM friendly
+
+    @staticmethod
+    def forward(ctx, x, weight, labels, ignore_index):
+        ext = _require_ext("cross_entropy")
+        e = x.shape[-1]
+        x2d = x.contiguous().view(-1, e)
+        lab = labels.contiguous().view(-1)
+        n = x2d.shape[0]
+        dx = torch.empty_like(x2d)
+        dw = torch.zeros_like(weight, dtype=torch.float32)
+        loss_sum = torch.zeros((), device=x.device, dtype=torch.float32)
+        count = (lab != ignore_index).sum()
+        denom = count.clamp(min=1).float()
+        for i in range(0, n, _LinearCEFn.CHUNK):
+            xc = x2d[i:i + _LinearCEFn.CHUNK]
+            lc = lab[i:i + _LinearCEFn.CHUNK]
+            logits = torch.mm(xc, weight.t())          # hipBLASLt GEMM, bf16
+            # in-place: logits -> dlogits (softmax - onehot)/denom, adds loss
+            ext.ce_fwd_bwd(logits, lc, loss_sum, denom, ignore_index)
+            torch.mm(logits, weight, out=dx[i:i + _LinearCEFn.CHUNK])
+            dw.add_(torch.mm(logits.t(), xc).float())
+        ctx.save_for_backward(dx, dw)
+        ctx.xshape = x.shape
+        ctx.wdtype = weight.dtype
+        return loss_sum / denom
+
+    @staticmethod
+    def backward(ctx, gout):
+        dx, dw = ctx.saved_tensors
+        return (dx.view(ctx.xshape) * gout, (dw * gout).to(ctx.wdtype),
+                None, None)
+
+
+def linear_cross_entropy(x, weight, labels, ignore_index=-100):
+    if x.is_cuda:
+        return _LinearCEFn.apply(x, weight, labels, ignore_index)
+    return reference.linear_cross_entropy(x, weight, labels, ignore_index)
+
+
+# --------------------------------------------------------------------------
+# Fused AdamW on flat fp32 shards + multi-tensor sq-norm
+# --------------------------------------------------------------------------
+def fused_adamw(p, g, m, v, step, lr, beta1, beta2, eps, weight_decay):
+    """In-place AdamW on 1-D fp32 tensors. g may be fp32 or bf16."""
+    if p.is_cuda:
+        ext = _require_ext("adamw")
+        ext.adamw(p, g, m, v, float(step), lr, beta1, beta2, eps, weight_decay)
+    else:
+        reference.adamw_step(p, g.float(), m, v, step, lr, beta1, beta2, eps,
+                             weight_decay)
+
+
+def sq_norm(tensors):
+    """Sum of squares over a list of tensors -> fp32 scalar tensor."""
+    if tensors and tensors[0].is_cuda and _C is not None:
+        out = torch.zeros((), device=tensors[0].device, dtype=torch.float32)
+        for t in tensors:
+            _C.sq_norm_accum(t.view(-1), out)
+        return out
+    return sum((t.float().pow(2).sum() for t in tensors),
+               torch.zeros((), dtype=torch.float32,
+                           device=tensors[0].device if tensors else "cpu"))

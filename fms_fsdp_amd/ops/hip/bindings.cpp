@@ -1,0 +1,183 @@
+// Python bindings for the CDNA4 kernel library (fms_fsdp_amd._C).
+// Pure dispatch: shape/dtype checks + launch on the current HIP stream.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime_api.h>
+
+#include <tuple>
+
+using torch::Tensor;
+
+extern "C" {
+void launch_rmsnorm_fwd(const void*, const void*, void*, float*, int, int,
+                        float, hipStream_t);
+void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
+                        void*, float*, int, int, hipStream_t);
+void launch_rope(const void*, void*, const float*, const float*, int, int,
+                 int, int, int, hipStream_t);
+void launch_swiglu_fwd(const void*, void*, long long, int, hipStream_t);
+void launch_swiglu_bwd(const void*, const void*, void*, long long, int,
+                       hipStream_t);
+void launch_ce_fwd_bwd(void*, const long long*, float*, const float*,
+                       long long, int, int, hipStream_t);
+void launch_adamw(float*, const void*, int, float*, float*, long long, float,
+                  float, float, float, float, float, float, hipStream_t);
+void launch_sqnorm(const void*, int, float*, long long, hipStream_t);
+void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
+                     int, int, int, int, int, float, hipStream_t);
+void launch_attn_bwd(const void*, const void*, const void*, const void*,
+                     const void*, const float*, void*, void*, void*, float*,
+                     int, int, int, int, int, float, hipStream_t);
+}
+
+static hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_BF16_CONTIG(t)                                        \
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, #t " must be bf16"); \
+  TORCH_CHECK(t.is_contiguous(), #t " must be contiguous");
+
+std::tuple<Tensor, Tensor> rmsnorm_fwd(Tensor x, Tensor w, double eps) {
+  CHECK_BF16_CONTIG(x);
+  CHECK_BF16_CONTIG(w);
+  const int rows = x.size(0), H = x.size(1);
+  TORCH_CHECK(H % 8 == 0, "H must be divisible by 8");
+  auto y = torch::empty_like(x);
+  auto rinv = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  launch_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                     rinv.data_ptr<float>(), rows, H, (float)eps,
+                     cur_stream());
+  return {y, rinv};
+}
+
+std::tuple<Tensor, Tensor> rmsnorm_bwd(Tensor dy, Tensor x, Tensor w,
+                                       Tensor rinv) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(x);
+  const int rows = x.size(0), H = x.size(1);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  launch_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                     rinv.data_ptr<float>(), dx.data_ptr(),
+                     dw.data_ptr<float>(), rows, H, cur_stream());
+  return {dx, dw};
+}
+
+std::tuple<Tensor, Tensor> rope_fwd(Tensor q, Tensor k, Tensor cos,
+                                    Tensor sin, bool conj) {
+  CHECK_BF16_CONTIG(q);
+  CHECK_BF16_CONTIG(k);
+  TORCH_CHECK(cos.scalar_type() == torch::kFloat32 && cos.is_contiguous());
+  const int b = q.size(0), s = q.size(1), h = q.size(2), d = q.size(3);
+  const int kvh = k.size(2);
+  TORCH_CHECK((d / 2) % 4 == 0, "head_dim/2 must be divisible by 4");
+  auto qo = torch::empty_like(q);
+  auto ko = torch::empty_like(k);
+  launch_rope(q.data_ptr(), qo.data_ptr(), cos.data_ptr<float>(),
+              sin.data_ptr<float>(), b, s, h, d, conj, cur_stream());
+  launch_rope(k.data_ptr(), ko.data_ptr(), cos.data_ptr<float>(),
+              sin.data_ptr<float>(), b, s, kvh, d, conj, cur_stream());
+  return {qo, ko};
+}
+
+Tensor swiglu_fwd(Tensor gu) {
+  CHECK_BF16_CONTIG(gu);
+  const long long rows = gu.size(0);
+  const int H2 = gu.size(1);
+  TORCH_CHECK(H2 % 16 == 0);
+  auto h = torch::empty({rows, H2 / 2}, gu.options());
+  launch_swiglu_fwd(gu.data_ptr(), h.data_ptr(), rows, H2 / 2, cur_stream());
+  return h;
+}
+
+Tensor swiglu_bwd(Tensor dy, Tensor gu) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(gu);
+  const long long rows = gu.size(0);
+  const int H2 = gu.size(1);
+  auto dgu = torch::empty_like(gu);
+  launch_swiglu_bwd(dy.data_ptr(), gu.data_ptr(), dgu.data_ptr(), rows,
+                    H2 / 2, cur_stream());
+  return dgu;
+}
+
+void ce_fwd_bwd(Tensor logits, Tensor labels, Tensor loss_sum, Tensor denom,
+                int64_t ignore_index) {
+  CHECK_BF16_CONTIG(logits);
+  TORCH_CHECK(labels.scalar_type() == torch::kInt64 && labels.is_contiguous());
+  const int rows = logits.size(0), V = logits.size(1);
+  launch_ce_fwd_bwd(logits.data_ptr(),
+                    reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
+                    loss_sum.data_ptr<float>(), denom.data_ptr<float>(),
+                    (long long)ignore_index, rows, V, cur_stream());
+}
+
+void adamw(Tensor p, Tensor g, Tensor m, Tensor v, double step, double lr,
+           double b1, double b2, double eps, double wd) {
+  TORCH_CHECK(p.scalar_type() == torch::kFloat32 && p.is_contiguous());
+  const long long n = p.numel();
+  TORCH_CHECK(n % 4 == 0, "shard size must be divisible by 4");
+  const bool gbf = g.scalar_type() == torch::kBFloat16;
+  const float bc1 = 1.f - powf((float)b1, (float)step);
+  const float bc2 = 1.f - powf((float)b2, (float)step);
+  launch_adamw(p.data_ptr<float>(), g.data_ptr(), gbf, m.data_ptr<float>(),
+               v.data_ptr<float>(), n, (float)lr, (float)b1, (float)b2,
+               (float)eps, (float)wd, bc1, bc2, cur_stream());
+}
+
+void sq_norm_accum(Tensor t, Tensor out) {
+  TORCH_CHECK(t.is_contiguous());
+  const long long n = t.numel();
+  TORCH_CHECK(n % 4 == 0);
+  launch_sqnorm(t.data_ptr(), t.scalar_type() == torch::kBFloat16,
+                out.data_ptr<float>(), n, cur_stream());
+}
+
+std::tuple<Tensor, Tensor> attn_fwd(Tensor q, Tensor k, Tensor v) {
+  CHECK_BF16_CONTIG(q);
+  CHECK_BF16_CONTIG(k);
+  CHECK_BF16_CONTIG(v);
+  const int b = q.size(0), s = q.size(1), h = q.size(2), d = q.size(3);
+  const int kvh = k.size(2);
+  TORCH_CHECK(d == 64 || d == 128, "head_dim must be 64 or 128");
+  TORCH_CHECK(s % 128 == 0, "seq len must be a multiple of 128");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({b, h, s}, q.options().dtype(torch::kFloat32));
+  const float scale = 1.f / sqrtf((float)d);
+  launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  lse.data_ptr<float>(), b, s, h, kvh, d, scale,
+                  cur_stream());
+  return {o, lse};
+}
+
+std::tuple<Tensor, Tensor, Tensor> attn_bwd(Tensor do_, Tensor q, Tensor k,
+                                            Tensor v, Tensor o, Tensor lse) {
+  CHECK_BF16_CONTIG(do_);
+  const int b = q.size(0), s = q.size(1), h = q.size(2), d = q.size(3);
+  const int kvh = k.size(2);
+  auto dq = torch::empty_like(q);
+  // dk/dv accumulated in fp32 (GQA head-groups collide), cast after
+  auto dk = torch::zeros({b, s, kvh, d}, q.options().dtype(torch::kFloat32));
+  auto dv = torch::zeros({b, s, kvh, d}, q.options().dtype(torch::kFloat32));
+  auto delta = torch::empty({b, h, s}, q.options().dtype(torch::kFloat32));
+  const float scale = 1.f / sqrtf((float)d);
+  launch_attn_bwd(do_.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                  o.data_ptr(), lse.data_ptr<float>(), dq.data_ptr(),
+                  dk.data_ptr(), dv.data_ptr(), delta.data_ptr<float>(), b, s,
+                  h, kvh, d, scale, cur_stream());
+  return {dq, dk.to(torch::kBFloat16), dv.to(torch::kBFloat16)};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("rmsnorm_fwd", &rmsnorm_fwd);
+  mod.def("rmsnorm_bwd", &rmsnorm_bwd);
+  mod.def("rope_fwd", &rope_fwd);
+  mod.def("swiglu_fwd", &swiglu_fwd);
+  mod.def("swiglu_bwd", &swiglu_bwd);
+  mod.def("ce_fwd_bwd", &ce_fwd_bwd);
+  mod.def("adamw", &adamw);
+  mod.def("sq_norm_accum", &sq_norm_accum);
+  mod.def("attn_fwd", &attn_fwd);
+  mod.def("attn_bwd", &attn_bwd);
+}

@@ -1,0 +1,472 @@
+"""MI355X-native sharded-training runtime (the FSDP/HSDP/DDP engine).
+
+Re-implements the capability of torch-FSDP as used by the reference
+(FULL_SHARD / HYBRID_SHARD / NO_SHARD, per-block wrapping, bf16 mixed
+precision with fp32 master shards, meta-device init — reference call sites:
+main_training_llama.py:82-91, fms_fsdp/policies/*), re-designed for one
+8-GPU xGMI clique per node with 288 GB HBM3E per GPU:
+
+- One flat bf16 parameter buffer per transformer block ("unit"); the local
+  shard is 1/S of it (S = shard-group size). all_gather_into_tensor over
+  RCCL reconstructs the full buffer on a dedicated HIP comm stream with
+  configurable lookahead prefetch, overlapped with the previous block's
+  compute (SURVEY.md §3.2 hot loop).
+- 288 GB HBM lets us default to reshard_after_forward=False for <=13B
+  models: parameters stay gathered between forward and backward, so each
+  step costs ONE all-gather + ONE reduce-scatter per unit instead of the
+  reference's two gathers (torch FSDP re-gathers in backward;
+  SURVEY.md §2.3 collectives table).
+- Gradients accumulate into a flat bf16 buffer (param.grad views);
+  when a unit's backward completes, a bf16 reduce_scatter_tensor runs on a
+  second stream and the result is accumulated into an fp32 shard for the
+  fused-AdamW HIP kernel. HSDP adds an inter-node all-reduce of the
+  reduce-scattered shard.
+- Optimizer state (fp32 master + AdamW moments) lives only on the shard.
+
+Collectives go through torch.distributed's "nccl" backend (= RCCL on ROCm)
+over the fully-connected 7-link xGMI clique; message sizes are whole-block
+shards (≈50 MB for 7B at S=8), large enough to saturate multi-ring RCCL.
+On CPU the same code runs on gloo (world_size>1 tested in tests/).
+"""
+
+import os
+from contextlib import nullcontext
+from typing import List, Optional, Type
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from fms_fsdp_amd import ops
+
+_ALIGN = 128  # element alignment for each param inside the flat buffer
+
+
+def _pad_to(n, k):
+    return (n + k - 1) // k * k
+
+
+class FlatUnit:
+    """One sharding unit: a module subtree whose params live in one flat
+    bf16 buffer, sharded 1/S per rank."""
+
+    def __init__(self, name: str, module: nn.Module, params: List[nn.Parameter],
+                 shard_group, replicate_group, device, param_dtype,
+                 reshard_after_forward: bool, param_names=None):
+        self.name = name
+        self.module = module
+        self.params = params
+        self.param_names = param_names or [f"p{i}" for i in range(len(params))]
+        self.shard_group = shard_group            # None => no sharding (S=1)
+        self.replicate_group = replicate_group    # None => no replication group
+        self.device = device
+        self.param_dtype = param_dtype
+        self.reshard_after_forward = reshard_after_forward
+        self.S = dist.get_world_size(shard_group) if shard_group is not None else 1
+
+        # layout
+        self.offsets = []
+        off = 0
+        for p in params:
+            self.offsets.append(off)
+            off += _pad_to(p.numel(), _ALIGN)
+        self.total = _pad_to(off, _ALIGN * max(self.S, 1))
+        self.shard_size = self.total // self.S
+
+        # persistent tensors (allocated in materialize())
+        self.flat_param: Optional[torch.Tensor] = None     # (total,) bf16
+        self.flat_grad: Optional[torch.Tensor] = None      # (total,) bf16
+        self.param_shard: Optional[torch.Tensor] = None    # (shard,) bf16
+        self.master_shard: Optional[torch.Tensor] = None   # (shard,) fp32
+        self.grad_shard: Optional[torch.Tensor] = None     # (shard,) fp32
+        self.exp_avg: Optional[torch.Tensor] = None        # fp32 (adamw m)
+        self.exp_avg_sq: Optional[torch.Tensor] = None     # fp32 (adamw v)
+
+        # runtime state
+        self._gathered = False
+        self._gather_event = None
+        self._grad_event = None
+        self._grad_countdown = 0
+        self._grads_ready_views = False
+
+    # ---------------- construction ----------------
+
+    def materialize(self, src_rank_broadcast: bool):
+        """Create shards from the module's current (initialized) params and
+        swap param storage to views of the flat buffer."""
+        dev = self.device
+        with torch.no_grad():
+            flat = torch.zeros(self.total, dtype=self.param_dtype, device=dev)
+            for p, off in zip(self.params, self.offsets):
+                flat[off:off + p.numel()].copy_(p.detach().reshape(-1).to(dev))
+            if src_rank_broadcast and dist.is_initialized() and dist.get_world_size() > 1:
+                dist.broadcast(flat, src=0)
+            rank = dist.get_rank(self.shard_group) if self.shard_group is not None else 0
+            if self.S > 1:
+                self.param_shard = flat[rank * self.shard_size:(rank + 1) * self.shard_size].clone()
+            self.flat_param = flat
+            if self.S == 1:
+                self.param_shard = self.flat_param  # same storage
+            self.master_shard = self.param_shard.float()
+            self.grad_shard = torch.zeros_like(self.master_shard)
+            self.exp_avg = torch.zeros_like(self.master_shard)
+            self.exp_avg_sq = torch.zeros_like(self.master_shard)
+            self.flat_grad = torch.zeros(self.total, dtype=self.param_dtype, device=dev)
+            self._point_params_into_flat()
+            self._set_grad_views()
+            self._gathered = True
+
+    def _point_params_into_flat(self):
+        with torch.no_grad():
+            for p, off in zip(self.params, self.offsets):
+                p.data = self.flat_param[off:off + p.numel()].view(p.shape)
+
+    def _set_grad_views(self):
+        for p, off in zip(self.params, self.offsets):
+            p.grad = self.flat_grad[off:off + p.numel()].view(p.shape)
+        self._grads_ready_views = True
+
+    # ---------------- storage control ----------------
+
+    def _free_flat_param(self):
+        if self.S == 1:
+            return  # flat IS the shard; never free
+        self.flat_param.untyped_storage().resize_(0)
+        self._gathered = False
+
+    def _alloc_flat_param(self):
+        st = self.flat_param.untyped_storage()
+        if st.size() == 0:
+            st.resize_(self.total * self.flat_param.element_size())
+
+    # ---------------- collectives ----------------
+
+    def gather(self, comm_stream):
+        """Issue async all-gather of the bf16 shard on the comm stream."""
+        if self._gathered:
+            return
+        if self.S == 1:
+            self._gathered = True
+            return
+        ctx = torch.cuda.stream(comm_stream) if comm_stream is not None else nullcontext()
+        with ctx:
+            self._alloc_flat_param()
+            dist.all_gather_into_tensor(self.flat_param, self.param_shard,
+                                        group=self.shard_group)
+            if comm_stream is not None:
+                self._gather_event = torch.cuda.Event()
+                self._gather_event.record(comm_stream)
+        self._gathered = True
+        self._point_params_into_flat()
+
+    def wait_gather(self):
+        if self._gather_event is not None:
+            torch.cuda.current_stream().wait_event(self._gather_event)
+            self.flat_param.record_stream(torch.cuda.current_stream())
+            self._gather_event = None
+
+    def mark_stale(self):
+        """After optimizer step the gathered buffers hold old weights."""
+        if self.S > 1:
+            if self.reshard_after_forward:
+                self._free_flat_param()
+            else:
+                self._gathered = False  # keep storage, re-gather into it
+
+    def reduce_grads(self, rs_stream, reduce_dtype):
+        """reduce-scatter flat bf16 grads -> fp32 grad_shard (+= for grad
+        accumulation); HSDP: all-reduce the shard across replicas."""
+        if rs_stream is not None:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            rs_stream.wait_event(ev)
+        rep = dist.get_world_size(self.replicate_group) if self.replicate_group is not None else 1
+        total_dp = self.S * rep  # grads averaged over ALL data-parallel ranks
+        ctx = torch.cuda.stream(rs_stream) if rs_stream is not None else nullcontext()
+        with ctx:
+            if self.S > 1:
+                rs_out = torch.empty(self.shard_size, dtype=self.flat_grad.dtype,
+                                     device=self.device)
+                dist.reduce_scatter_tensor(rs_out, self.flat_grad,
+                                           op=dist.ReduceOp.SUM, group=self.shard_group)
+                rsf = rs_out.float()
+                if total_dp > 1:
+                    rsf /= total_dp
+                if self.replicate_group is not None:
+                    dist.all_reduce(rsf, group=self.replicate_group)
+                self.grad_shard.add_(rsf)
+                del rs_out
+            else:
+                rsf = self.flat_grad.float()
+                if total_dp > 1:
+                    rsf /= total_dp
+                if self.replicate_group is not None:
+                    dist.all_reduce(rsf, group=self.replicate_group)
+                self.grad_shard.add_(rsf)
+            # grads for this unit are consumed; zero the bf16 buffer for the
+            # next backward (accumulation now lives in fp32 grad_shard)
+            self.flat_grad.zero_()
+            if rs_stream is not None:
+                self._grad_event = torch.cuda.Event()
+                self._grad_event.record(rs_stream)
+
+    def wait_grads(self):
+        if self._grad_event is not None:
+            torch.cuda.current_stream().wait_event(self._grad_event)
+            self._grad_event = None
+
+    # ---------------- optimizer plumbing ----------------
+
+    def publish_master_to_shard(self):
+        """master fp32 -> bf16 param shard (source of next all-gather)."""
+        with torch.no_grad():
+            self.param_shard.copy_(self.master_shard)
+
+    def sharded_numel_unpadded(self):
+        return sum(p.numel() for p in self.params)
+
+
+class ShardedModel(nn.Module):
+    """Wraps a model; shards every `block_class` submodule as a unit plus one
+    root unit for the remaining params (embedding / final norm / lm_head)."""
+
+    def __init__(self, model: nn.Module, block_class: Type[nn.Module],
+                 sharding_strategy: str = "fsdp",
+                 device=None,
+                 param_dtype=torch.bfloat16,
+                 reduce_dtype=torch.bfloat16,
+                 reshard_after_forward: bool = False,
+                 prefetch_lookahead: int = 1,
+                 intra_node_size: Optional[int] = None):
+        super().__init__()
+        self.model = model
+        self.prefetch_lookahead = max(prefetch_lookahead, 0)
+        self.reduce_dtype = reduce_dtype
+        self._step_trained = False
+
+        world = dist.get_world_size() if dist.is_initialized() else 1
+        rank = dist.get_rank() if dist.is_initialized() else 0
+        if device is None:
+            device = torch.device("cuda", torch.cuda.current_device()) \
+                if torch.cuda.is_available() else torch.device("cpu")
+        self.device = device
+        self.is_cuda = device.type == "cuda"
+
+        # process groups per strategy
+        shard_group = replicate_group = None
+        if world > 1:
+            if sharding_strategy == "fsdp":
+                shard_group = dist.group.WORLD
+            elif sharding_strategy == "hsdp":
+                intra = intra_node_size or int(os.environ.get("LOCAL_WORLD_SIZE", 0)) \
+                    or min(world, torch.cuda.device_count() or world)
+                if world % intra != 0:
+                    intra = world
+                mesh_shard, mesh_rep = _build_hsdp_groups(world, rank, intra)
+                shard_group, replicate_group = mesh_shard, mesh_rep
+                if dist.get_world_size(shard_group) == 1:
+                    shard_group = None
+            elif sharding_strategy == "ddp":
+                shard_group = None
+                replicate_group = dist.group.WORLD
+            else:
+                raise ValueError(f"unknown sharding strategy {sharding_strategy}")
+        self.shard_group = shard_group
+        self.replicate_group = replicate_group
+
+        # build units: one per block, one root
+        blocks = [(n, m) for n, m in model.named_modules() if isinstance(m, block_class)]
+        name_of = {id(p): n for n, p in model.named_parameters()}
+        block_param_ids = set()
+        self.units: List[FlatUnit] = []
+        for n, m in blocks:
+            params = [p for p in m.parameters() if p.requires_grad]
+            block_param_ids.update(id(p) for p in params)
+            self.units.append(FlatUnit(n, m, params, shard_group, replicate_group,
+                                       device, param_dtype, reshard_after_forward,
+                                       param_names=[name_of[id(p)] for p in params]))
+        root_params = [p for p in model.parameters()
+                       if p.requires_grad and id(p) not in block_param_ids]
+        self.root_unit = FlatUnit("_root", model, root_params, shard_group,
+                                  replicate_group, device, param_dtype, False,
+                                  param_names=[name_of[id(p)] for p in root_params])
+        self.all_units = [self.root_unit] + self.units
+        self._unit_of_module = {id(u.module): u for u in self.units}
+
+        # move buffers (e.g. rope tables) to device, keep dtype
+        for b in model.buffers():
+            b.data = b.data.to(device)
+
+        # materialize shards (broadcast rank-0 init for determinism)
+        for u in self.all_units:
+            u.materialize(src_rank_broadcast=True)
+
+        # streams
+        self.comm_stream = torch.cuda.Stream() if self.is_cuda else None
+        self.rs_stream = torch.cuda.Stream() if self.is_cuda else None
+
+        self._install_hooks()
+
+    # ---------------- hooks / orchestration ----------------
+
+    def _install_hooks(self):
+        for i, u in enumerate(self.units):
+            u.module.register_forward_pre_hook(self._make_fwd_pre(i))
+            u.module.register_forward_hook(self._make_fwd_post(i))
+            u.module.register_full_backward_pre_hook(self._make_bwd_pre(i))
+        for u in self.all_units:
+            for p in u.params:
+                p.register_post_accumulate_grad_hook(self._make_grad_hook(u))
+
+    def _make_fwd_pre(self, idx):
+        def hook(module, args):
+            u = self.units[idx]
+            u.gather(self.comm_stream)
+            u.wait_gather()
+            for j in range(idx + 1, min(idx + 1 + self.prefetch_lookahead,
+                                        len(self.units))):
+                self.units[j].gather(self.comm_stream)
+        return hook
+
+    def _make_fwd_post(self, idx):
+        def hook(module, args, out):
+            u = self.units[idx]
+            if u.reshard_after_forward and torch.is_grad_enabled():
+                u._free_flat_param()
+            return None
+        return hook
+
+    def _make_bwd_pre(self, idx):
+        def hook(module, grad_output):
+            u = self.units[idx]
+            if not u._gathered:
+                u.gather(self.comm_stream)
+            u.wait_gather()
+            # prefetch previous blocks (backward order)
+            for j in range(idx - 1, max(idx - 1 - self.prefetch_lookahead, -1), -1):
+                if self.units[j].reshard_after_forward:
+                    self.units[j].gather(self.comm_stream)
+            if u._grad_countdown == 0:
+                u._grad_countdown = len(u.params)
+            return None
+        return hook
+
+    def _make_grad_hook(self, u: FlatUnit):
+        def hook(param):
+            if u._grad_countdown == 0:
+                u._grad_countdown = len(u.params)
+            u._grad_countdown -= 1
+            if u._grad_countdown == 0:
+                u.reduce_grads(self.rs_stream, self.reduce_dtype)
+                if u is not self.root_unit and u.reshard_after_forward:
+                    u._free_flat_param()
+        return hook
+
+    def forward(self, *args, **kwargs):
+        # root params (embedding) needed first; gather root + lookahead
+        self.root_unit.gather(self.comm_stream)
+        self.root_unit.wait_gather()
+        for j in range(min(self.prefetch_lookahead, len(self.units))):
+            self.units[j].gather(self.comm_stream)
+        return self.model(*args, **kwargs)
+
+    # ---------------- training utilities ----------------
+
+    def clip_grad_norm_(self, max_norm):
+        """Global grad-norm over fp32 grad shards + clip (reference:
+        train_utils.py:96 model.clip_grad_norm_)."""
+        for u in self.all_units:
+            u.wait_grads()
+        local = ops.sq_norm([u.grad_shard for u in self.all_units])
+        if self.shard_group is not None:
+            dist.all_reduce(local, group=self.shard_group)
+        total_norm = local.sqrt()
+        clip = max_norm / (total_norm + 1e-6)
+        if self.is_cuda:
+            coef = torch.clamp(clip, max=1.0)
+            for u in self.all_units:
+                u.grad_shard.mul_(coef)
+        else:
+            c = min(clip.item(), 1.0)
+            if c < 1.0:
+                for u in self.all_units:
+                    u.grad_shard.mul_(c)
+        return total_norm
+
+    def zero_grad(self, set_to_none=False):
+        for u in self.all_units:
+            u.grad_shard.zero_()
+            if not u._grads_ready_views:
+                u._set_grad_views()
+
+    def param_count(self):
+        return sum(u.sharded_numel_unpadded() for u in self.all_units)
+
+
+class ShardedAdamW:
+    """Fused AdamW over the fp32 master shards (HIP multi-tensor kernel on
+    GPU; reference hyperparams main_training_llama.py:113-115)."""
+
+    def __init__(self, sharded_model: ShardedModel, lr=3e-4, betas=(0.9, 0.95),
+                 eps=1e-8, weight_decay=0.1):
+        self.m = sharded_model
+        self.lr = lr
+        self.betas = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.step_count = 0
+        # LambdaLR compatibility
+        self.param_groups = [{"lr": lr, "initial_lr": lr}]
+        self.defaults = {"lr": lr}
+
+    def step(self):
+        self.step_count += 1
+        lr = self.param_groups[0]["lr"]
+        b1, b2 = self.betas
+        for u in self.m.all_units:
+            u.wait_grads()
+            ops.fused_adamw(u.master_shard, u.grad_shard, u.exp_avg,
+                            u.exp_avg_sq, self.step_count, lr, b1, b2,
+                            self.eps, self.weight_decay)
+            u.publish_master_to_shard()
+            u.mark_stale()
+
+    def zero_grad(self, set_to_none=False):
+        self.m.zero_grad()
+
+    def state_dict(self):
+        return {
+            "step": self.step_count,
+            "lr": self.param_groups[0]["lr"],
+            "units": {u.name: {"exp_avg": u.exp_avg, "exp_avg_sq": u.exp_avg_sq,
+                               "master": u.master_shard}
+                      for u in self.m.all_units},
+        }
+
+    def load_state_dict(self, sd):
+        self.step_count = sd["step"]
+        self.param_groups[0]["lr"] = sd["lr"]
+        for u in self.m.all_units:
+            usd = sd["units"][u.name]
+            u.exp_avg.copy_(usd["exp_avg"])
+            u.exp_avg_sq.copy_(usd["exp_avg_sq"])
+            u.master_shard.copy_(usd["master"])
+            u.publish_master_to_shard()
+            u.mark_stale()
+
+
+def _build_hsdp_groups(world, rank, intra):
+    """Shard intra-node (xGMI clique), replicate inter-node."""
+    n_nodes = world // intra
+    shard_group = replicate_group = None
+    for node in range(n_nodes):
+        ranks = list(range(node * intra, (node + 1) * intra))
+        g = dist.new_group(ranks)
+        if rank in ranks:
+            shard_group = g
+    for local in range(intra):
+        ranks = list(range(local, world, intra))
+        g = dist.new_group(ranks)
+        if rank in ranks:
+            replicate_group = g
+    return shard_group, replicate_group

@@ -1,0 +1,294 @@
+"""Sharded checkpointing for the MI355X runtime.
+
+Parity target: reference fms_fsdp/utils/checkpointing_utils.py:23-316 —
+sharded folder checkpoints with HSDP write-dedup (only replicate-rank-0
+writes model/optim; every rank writes loader state), newest-n retention of
+"tmp" checkpoints, auto-discovery of the latest valid checkpoint,
+fresh-start vs job-resume vs continued-pretraining precedence, metadata
+{step, tokens_seen}, and single-file .pth load.
+
+Format (ours — the flat-shard layout of parallel/fsdp.py instead of torch
+DCP, self-describing for resharding and for the HF exporters):
+  step_<N>_ckp/
+    model_<r>_of_<S>.pth   {unit_name: fp32 master shard}
+    optim_<r>_of_<S>.pth   {unit_name: {exp_avg, exp_avg_sq}, step, lr}
+    metadata.pth           {step, tokens_seen, shard_world, layout:
+                            {unit: {names, shapes, offsets, total}}}
+    loader_state_<rank>.pth
+Shard files round-trip across world sizes: per-param offsets depend only on
+the 128-element alignment, not S, so load() reconstructs each unit's flat
+buffer from the old shard set and re-slices for the new S.
+"""
+
+import os
+import shutil
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def get_latest(targdir, qualifier=lambda x: True, key="step"):
+    """Latest checkpoint folder/file in targdir by step number
+    (reference: checkpointing_utils.py:23-41)."""
+    if not os.path.exists(targdir):
+        return targdir
+    latest = targdir
+    best = -1
+    for fi in os.listdir(targdir):
+        full = os.path.join(targdir, fi)
+        if key in fi and qualifier(full):
+            try:
+                num = int("".join(c for c in fi.split(key)[1].split("_")[1]
+                                  if c.isdigit())) if False else \
+                    int(fi.replace("_ckp", "").split("_")[-1])
+            except ValueError:
+                continue
+            if num > best:
+                best = num
+                latest = full
+    return latest
+
+
+def get_oldest(targdir, qualifier=lambda x: True):
+    """Oldest checkpoint by ctime (reference: checkpointing_utils.py:44-62)."""
+    oldest = targdir
+    best = float("inf")
+    if not os.path.exists(targdir):
+        return targdir
+    for fi in os.listdir(targdir):
+        full = os.path.join(targdir, fi)
+        if qualifier(full):
+            t = os.path.getctime(full)
+            if t < best:
+                best = t
+                oldest = full
+    return oldest
+
+
+class Checkpointer:
+    """Manages save/load of model+optimizer flat shards, dataloader state
+    and metadata (reference Checkpointer: checkpointing_utils.py:65-316)."""
+
+    def __init__(self, ckpt_dir, n_to_save, parallel_mode, rank, local_rank,
+                 report_fn=None):
+        self.max_ckps = n_to_save
+        self.rank = rank
+        self.local_rank = local_rank
+        self.ckpt_dir = ckpt_dir
+        assert parallel_mode in ("fsdp", "hsdp", "ddp")
+        self.parallel_mode = parallel_mode
+        self.report = report_fn if report_fn is not None else self._default_report
+
+    def _default_report(self, output_path=None, **kwargs):
+        if self.rank == 0:
+            if output_path is not None:
+                print(f"Checkpoint saved to {output_path}")
+            for k, v in kwargs.items():
+                print(f"{k}: {v}")
+
+    # ------------- helpers -------------
+
+    @staticmethod
+    def _shard_info(model):
+        """(shard_rank, S, is_writer) from the ShardedModel's groups."""
+        sg = model.shard_group
+        rg = model.replicate_group
+        shard_rank = dist.get_rank(sg) if sg is not None else 0
+        S = dist.get_world_size(sg) if sg is not None else 1
+        rep_rank = dist.get_rank(rg) if rg is not None else 0
+        return shard_rank, S, rep_rank == 0
+
+    @staticmethod
+    def _layout(model):
+        return {u.name: {"names": u.param_names,
+                         "shapes": [list(p.shape) for p in u.params],
+                         "offsets": u.offsets,
+                         "total": u.total}
+                for u in model.all_units}
+
+    def _cleanup(self):
+        qual = lambda x: os.path.basename(x).endswith("_ckp")
+        files = [f for f in os.listdir(self.ckpt_dir)
+                 if qual(os.path.join(self.ckpt_dir, f))] \
+            if os.path.exists(self.ckpt_dir) else []
+        if len(files) > self.max_ckps:
+            oldest = get_oldest(self.ckpt_dir, qualifier=qual)
+            if self.rank == 0 and oldest != self.ckpt_dir:
+                shutil.rmtree(oldest, ignore_errors=True)
+
+    def _validate_ckp_path(self, path):
+        """Return a valid checkpoint target under/at path, else None
+        (reference: checkpointing_utils.py:165-182)."""
+        if not os.path.exists(path):
+            return None
+        if os.path.isfile(path):
+            return path
+        if os.path.exists(os.path.join(path, "metadata.pth")):
+            return path
+        latest = get_latest(path)
+        if latest != path and os.path.exists(os.path.join(latest, "metadata.pth")):
+            return latest
+        if os.path.isfile(latest):
+            return latest
+        return None
+
+    # ------------- save -------------
+
+    def save(self, step, model, optimizer, dataloader, tokens_seen=0):
+        t0 = time.time()
+        shard_rank, S, is_writer = self._shard_info(model)
+        out = os.path.join(self.ckpt_dir, f"step_{step}_ckp")
+        os.makedirs(out, exist_ok=True)
+        if dist.is_initialized():
+            dist.barrier()
+        if is_writer:
+            model_sd = {u.name: u.master_shard.cpu() for u in model.all_units}
+            torch.save(model_sd, os.path.join(out, f"model_{shard_rank}_of_{S}.pth"))
+            opt_sd = {"step": optimizer.step_count, "lr": optimizer.param_groups[0]["lr"],
+                      "units": {u.name: {"exp_avg": u.exp_avg.cpu(),
+                                         "exp_avg_sq": u.exp_avg_sq.cpu()}
+                                for u in model.all_units}}
+            torch.save(opt_sd, os.path.join(out, f"optim_{shard_rank}_of_{S}.pth"))
+        if dataloader is not None and hasattr(dataloader.dataset, "state_dict"):
+            torch.save(dataloader.dataset.state_dict(),
+                       os.path.join(out, f"loader_state_{self.rank}.pth"))
+        if self.rank == 0:
+            torch.save({"step": step, "tokens_seen": tokens_seen,
+                        "shard_world": S, "layout": self._layout(model)},
+                       os.path.join(out, "metadata.pth"))
+        if dist.is_initialized():
+            dist.barrier()
+        self.report(output_path=out, time_taken=f"{time.time() - t0:.2f}s")
+        self._cleanup()
+        return out
+
+    # ------------- load -------------
+
+    def load(self, model, optimizer, dataloader, path="", reset_stepcount=False,
+             strict=True, is_compiled=False):
+        """Auto-discovering load. Prefers a checkpoint in the save dir (job
+        resume) over `path` (continued pretraining, step reset)
+        (reference: checkpointing_utils.py:184-281)."""
+        save_dir_ckpt = self._validate_ckp_path(self.ckpt_dir)
+        if save_dir_ckpt is not None:
+            load_path = save_dir_ckpt
+            reset_stepcount = False
+        else:
+            load_path = self._validate_ckp_path(path) if path else None
+            if load_path is None:
+                self.report(msg="No valid checkpoint detected, starting fresh")
+                return model, optimizer, dataloader, 0, 0, False
+            reset_stepcount = True
+        self.report(msg=f"Prior checkpoint {load_path} detected")
+
+        if os.path.isfile(load_path):
+            sd = torch.load(load_path, map_location="cpu", weights_only=False)
+            if "model_state" in sd:
+                sd = sd["model_state"]
+            sd = {k.replace("_orig_mod.", ""): v for k, v in sd.items()}
+            self._load_full_state_dict(model, sd, strict)
+            self.report(msg="Checkpoint loaded (single file, model only)",
+                        ckpt=load_path)
+            return model, optimizer, dataloader, 0, 0, False
+
+        meta = torch.load(os.path.join(load_path, "metadata.pth"),
+                          map_location="cpu", weights_only=False)
+        old_S = meta["shard_world"]
+        shard_rank, S, _ = self._shard_info(model)
+        self._load_model_shards(model, load_path, meta, shard_rank, S, old_S)
+        step = 0 if reset_stepcount else meta["step"]
+        tokens = 0 if reset_stepcount else meta.get("tokens_seen", 0)
+        if optimizer is not None and not reset_stepcount:
+            self._load_optim_shards(model, optimizer, load_path, meta,
+                                    shard_rank, S, old_S)
+        is_resuming = not reset_stepcount
+        if dataloader is not None and is_resuming and \
+                hasattr(dataloader.dataset, "load_state_dict"):
+            lp = os.path.join(load_path, f"loader_state_{self.rank}.pth")
+            if os.path.exists(lp):
+                dataloader.dataset.load_state_dict(
+                    torch.load(lp, map_location="cpu", weights_only=False))
+        self.report(msg=f"Checkpoint loaded from {load_path}", step=step)
+        return model, optimizer, dataloader, step, tokens, is_resuming
+
+    def _iter_unit_flat(self, load_path, meta, kind, key=None):
+        """Yield (unit_name, full flat fp32 tensor) reconstructed from the
+        old shard set (streamed one unit at a time)."""
+        old_S = meta["shard_world"]
+        shards = []
+        for r in range(old_S):
+            f = os.path.join(load_path, f"{kind}_{r}_of_{old_S}.pth")
+            shards.append(torch.load(f, map_location="cpu", weights_only=False))
+        for uname, info in meta["layout"].items():
+            if kind == "model":
+                parts = [shards[r][uname] for r in range(old_S)]
+            else:
+                parts = [shards[r]["units"][uname][key] for r in range(old_S)]
+            yield uname, torch.cat(parts)
+
+    def _load_model_shards(self, model, load_path, meta, shard_rank, S, old_S):
+        units = {u.name: u for u in model.all_units}
+        for uname, flat in self._iter_unit_flat(load_path, meta, "model"):
+            u = units[uname]
+            n = min(flat.numel(), u.total)
+            dst = torch.zeros(u.total, dtype=torch.float32)
+            dst[:n] = flat[:n]
+            shard = dst[shard_rank * u.shard_size:(shard_rank + 1) * u.shard_size]
+            u.master_shard.copy_(shard.to(u.master_shard.device))
+            u.publish_master_to_shard()
+            u.mark_stale()
+
+    def _load_optim_shards(self, model, optimizer, load_path, meta,
+                           shard_rank, S, old_S):
+        units = {u.name: u for u in model.all_units}
+        f0 = os.path.join(load_path, f"optim_0_of_{old_S}.pth")
+        if not os.path.exists(f0):
+            return
+        head = torch.load(f0, map_location="cpu", weights_only=False)
+        optimizer.step_count = head["step"]
+        optimizer.param_groups[0]["lr"] = head["lr"]
+        for key, attr in (("exp_avg", "exp_avg"), ("exp_avg_sq", "exp_avg_sq")):
+            for uname, flat in self._iter_unit_flat(load_path, meta, "optim", key):
+                u = units[uname]
+                n = min(flat.numel(), u.total)
+                dst = torch.zeros(u.total, dtype=torch.float32)
+                dst[:n] = flat[:n]
+                shard = dst[shard_rank * u.shard_size:(shard_rank + 1) * u.shard_size]
+                getattr(u, attr).copy_(shard.to(u.exp_avg.device))
+
+    def _load_full_state_dict(self, model, sd, strict):
+        """Load a name->tensor full state dict into the sharded model."""
+        shard_rank, S, _ = self._shard_info(model)
+        for u in model.all_units:
+            flat = torch.zeros(u.total, dtype=torch.float32)
+            for name, p, off in zip(u.param_names, u.params, u.offsets):
+                if name in sd:
+                    flat[off:off + p.numel()] = sd[name].reshape(-1).float()
+                elif strict:
+                    raise KeyError(f"missing key {name} in checkpoint")
+            shard = flat[shard_rank * u.shard_size:(shard_rank + 1) * u.shard_size]
+            u.master_shard.copy_(shard.to(u.master_shard.device))
+            u.publish_master_to_shard()
+            u.mark_stale()
+
+
+def consolidate_checkpoint(ckpt_path, dtype=torch.float32):
+    """Offline (no-dist) reconstruction of the full name->tensor state dict
+    from a sharded checkpoint folder — used by the fms_to_hf exporters
+    (reference analog: DCP no-dist load, fms_to_hf_llama.py:133-155)."""
+    meta = torch.load(os.path.join(ckpt_path, "metadata.pth"),
+                      map_location="cpu", weights_only=False)
+    old_S = meta["shard_world"]
+    shards = [torch.load(os.path.join(ckpt_path, f"model_{r}_of_{old_S}.pth"),
+                         map_location="cpu", weights_only=False)
+              for r in range(old_S)]
+    out = {}
+    for uname, info in meta["layout"].items():
+        flat = torch.cat([shards[r][uname] for r in range(old_S)])
+        for name, shape, off in zip(info["names"], info["shapes"], info["offsets"]):
+            n = 1
+            for s in shape:
+                n *= s
+            out[name] = flat[off:off + n].view(shape).to(dtype)
+    return out

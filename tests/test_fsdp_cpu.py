@@ -1,0 +1,186 @@
+"""Multi-process (gloo, world_size=2) tests of the sharded runtime —
+coverage the reference lacks entirely (SURVEY.md §4 'the new framework must
+add real distributed tests')."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+PORT = 29650
+
+
+def _single_process_reference(seed=0, steps=3, lr=1e-3):
+    from fms_fsdp_amd.models import Llama, LlamaBlock, LlamaConfig
+    from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+    torch.manual_seed(seed)
+    cfg = LlamaConfig(src_vocab_size=128, emb_dim=64, nheads=4, kvheads=2,
+                      nlayers=3, max_expected_seq_len=64)
+    m = Llama(cfg)
+    m.reset_parameters()
+    sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                      param_dtype=torch.float32)
+    opt = ShardedAdamW(sm, lr=lr)
+    g = torch.Generator().manual_seed(42)
+    x = torch.randint(0, 128, (2, 32), generator=g)
+    y = torch.randint(0, 128, (2, 32), generator=g)
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = sm(x, labels=y)
+        loss.backward()
+        sm.clip_grad_norm_(1.0)
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
+def _worker(rank, world, port, strat, reshard, q, steps=3):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from fms_fsdp_amd.models import Llama, LlamaBlock, LlamaConfig
+        from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+        torch.manual_seed(0)
+        cfg = LlamaConfig(src_vocab_size=128, emb_dim=64, nheads=4, kvheads=2,
+                          nlayers=3, max_expected_seq_len=64)
+        m = Llama(cfg)
+        m.reset_parameters()
+        sm = ShardedModel(m, LlamaBlock, sharding_strategy=strat,
+                          param_dtype=torch.float32, reshard_after_forward=reshard,
+                          intra_node_size=2)
+        opt = ShardedAdamW(sm, lr=1e-3)
+        g = torch.Generator().manual_seed(42)
+        x = torch.randint(0, 128, (2, 32), generator=g)
+        y = torch.randint(0, 128, (2, 32), generator=g)
+        losses = []
+        for _ in range(steps):
+            opt.zero_grad()
+            loss = sm(x, labels=y)
+            loss.backward()
+            sm.clip_grad_norm_(1.0)
+            opt.step()
+            losses.append(loss.item())
+        if rank == 0:
+            q.put(("ok", losses))
+    except Exception as e:  # pragma: no cover
+        q.put(("err", f"{type(e).__name__}: {e}"))
+        raise
+    finally:
+        dist.destroy_process_group()
+
+
+def _run_world2(strat, reshard, port):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_worker, args=(2, port, strat, reshard, q), nprocs=2, join=True)
+    status, payload = q.get()
+    assert status == "ok", payload
+    return payload
+
+
+@pytest.mark.parametrize("strat,reshard,port", [
+    ("fsdp", False, PORT + 1),
+    ("fsdp", True, PORT + 2),
+    ("hsdp", False, PORT + 3),
+    ("ddp", False, PORT + 4),
+])
+def test_world2_matches_single_process(strat, reshard, port):
+    """Identical batches on both ranks => identical training trajectory to
+    a single process (sharding must be numerically transparent)."""
+    ref = _single_process_reference()
+    got = _run_world2(strat, reshard, port)
+    for a, b in zip(ref, got):
+        assert abs(a - b) < 1e-5, (strat, ref, got)
+
+
+def _ckpt_worker(rank, world, port, tmpdir, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from fms_fsdp_amd.models import Llama, LlamaBlock, LlamaConfig
+        from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+        from fms_fsdp_amd.utils.checkpointing import Checkpointer
+        torch.manual_seed(0)
+        cfg = LlamaConfig(src_vocab_size=128, emb_dim=64, nheads=4, kvheads=2,
+                          nlayers=2, max_expected_seq_len=64)
+        m = Llama(cfg)
+        m.reset_parameters()
+        sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                          param_dtype=torch.float32)
+        opt = ShardedAdamW(sm, lr=1e-3)
+        g = torch.Generator().manual_seed(7)
+        x = torch.randint(0, 128, (2, 16), generator=g)
+        y = torch.randint(0, 128, (2, 16), generator=g)
+        for _ in range(2):
+            opt.zero_grad()
+            sm(x, labels=y).backward()
+            opt.step()
+        ck = Checkpointer(tmpdir, 3, "fsdp", rank, rank)
+        ck.save(2, sm, opt, None, tokens_seen=123)
+        if rank == 0:
+            torch.save([u.master_shard.clone() for u in sm.all_units],
+                       os.path.join(tmpdir, "rank0_shards.pth"))
+            q.put(("ok", None))
+    except Exception as e:
+        q.put(("err", f"{type(e).__name__}: {e}"))
+        raise
+    finally:
+        dist.destroy_process_group()
+
+
+def test_checkpoint_save_and_reshard_load(tmp_path):
+    """Save at world=2, reload single-process (world=1): master shards must
+    reassemble exactly (rescalable checkpoint, SURVEY.md hard-part 5)."""
+    tmpdir = str(tmp_path)
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_ckpt_worker, args=(2, PORT + 10, tmpdir, q), nprocs=2, join=True)
+    status, payload = q.get()
+    assert status == "ok", payload
+    rank0_shards = torch.load(os.path.join(tmpdir, "rank0_shards.pth"),
+                              weights_only=False)
+
+    from fms_fsdp_amd.models import Llama, LlamaBlock, LlamaConfig
+    from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+    from fms_fsdp_amd.utils.checkpointing import Checkpointer, get_latest
+    torch.manual_seed(0)
+    cfg = LlamaConfig(src_vocab_size=128, emb_dim=64, nheads=4, kvheads=2,
+                      nlayers=2, max_expected_seq_len=64)
+    m = Llama(cfg)
+    m.reset_parameters()
+    sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                      param_dtype=torch.float32)
+    opt = ShardedAdamW(sm, lr=1e-3)
+    ck = Checkpointer(tmpdir, 3, "fsdp", 0, 0)
+    _, _, _, step, tokens, resuming = ck.load(sm, opt, None, path="")
+    assert step == 2 and tokens == 123 and resuming
+    # world=1 master shard must equal the concatenation of the world=2 halves
+    for u, old0 in zip(sm.all_units, rank0_shards):
+        half = old0.numel()
+        assert torch.equal(u.master_shard[:half], old0), u.name
+
+
+def test_consolidate_checkpoint(tmp_path):
+    """consolidate_checkpoint reconstructs exact full params offline."""
+    from fms_fsdp_amd.models import Llama, LlamaBlock, LlamaConfig
+    from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+    from fms_fsdp_amd.utils.checkpointing import Checkpointer, consolidate_checkpoint
+    torch.manual_seed(3)
+    cfg = LlamaConfig(src_vocab_size=128, emb_dim=64, nheads=4, kvheads=2,
+                      nlayers=2, max_expected_seq_len=64)
+    m = Llama(cfg)
+    m.reset_parameters()
+    ref = {n: p.detach().clone() for n, p in m.named_parameters()}
+    sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                      param_dtype=torch.float32)
+    opt = ShardedAdamW(sm, lr=1e-3)
+    ck = Checkpointer(str(tmp_path), 3, "fsdp", 0, 0)
+    out = ck.save(1, sm, opt, None)
+    sd = consolidate_checkpoint(out)
+    for n, t in ref.items():
+        assert torch.allclose(sd[n], t.float(), atol=1e-6), n
