@@ -225,8 +225,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 __global__ void attn_bwd_delta_kernel(const short* __restrict__ dog,
                                       const short* __restrict__ og,
                                       float* __restrict__ delta,
-                                      int D, long long rows) {
-  // one wave per row
+                                      int D, int S, int H, long long rows) {
+  // one wave per row; input rows are (b, s, h) order, delta is (b, h, s)
   const long long row = ((long long)blockIdx.x * blockDim.x + threadIdx.x) / 64;
   if (row >= rows) return;
   const int lane = threadIdx.x & 63;
@@ -237,7 +237,12 @@ __global__ void attn_bwd_delta_kernel(const short* __restrict__ dog,
     acc += bf2f(dop[i]) * bf2f(op[i]) + bf2f(dop[i + 1]) * bf2f(op[i + 1]);
   }
   acc = wave_reduce_sum(acc);
-  if (lane == 0) delta[row] = acc;
+  if (lane == 0) {
+    const long long b = row / ((long long)S * H);
+    const int si = (int)((row / H) % S);
+    const int hi = (int)(row % H);
+    delta[(b * H + hi) * (long long)S + si] = acc;
+  }
 }
 
 // ---------------------------------------------------------------------
@@ -637,7 +642,7 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
                      int KVH, int D, float scale, hipStream_t stream) {
   const long long rows = (long long)B * S * H;
   attn_bwd_delta_kernel<<<(int)((rows * 64 + 255) / 256), 256, 0, stream>>>(
-      (const short*)do_, (const short*)o, delta_ws, D, rows);
+      (const short*)do_, (const short*)o, delta_ws, D, S, H, rows);
   dim3 grid(S / 128, B * H);
   if (D == 128) {
     const int lds_dq = 2 * 32 * 128 * 2 + 128 * 64;   // k + v + kt
