@@ -66,10 +66,17 @@ def main():
 
     bs, sl = args.batch_size, args.seq_len
     g = torch.Generator().manual_seed(1234 + rank)
-    inp = torch.randint(0, mcfg.src_vocab_size, (bs, sl + 1), generator=g).to(device)
-    x, y = inp[:, :-1], inp[:, 1:].contiguous()
+    # distinct synthetic batch per step (no single-batch memorization:
+    # the reported loss stays a meaningful ~log(V))
+    nb = 8
+    batches = [torch.randint(0, mcfg.src_vocab_size, (bs, sl + 1),
+                             generator=g).to(device) for _ in range(nb)]
+    it = [0]
 
     def step():
+        inp = batches[it[0] % nb]
+        it[0] += 1
+        x, y = inp[:, :-1], inp[:, 1:].contiguous()
         opt.zero_grad()
         loss = sm(x, labels=y)
         loss.backward()

@@ -191,6 +191,15 @@ def linear_cross_entropy(x, weight, labels, ignore_index=-100):
 
 
 # --------------------------------------------------------------------------
+# Causal depthwise conv1d + silu (mamba path). Runs through torch (MIOpen
+# depthwise conv) on GPU for now; a dedicated fused HIP kernel is on the
+# worklist (SURVEY.md §2.3 causal-conv1d row).
+# --------------------------------------------------------------------------
+def causal_conv1d(x, weight, bias):
+    return reference.causal_conv1d(x, weight, bias)
+
+
+# --------------------------------------------------------------------------
 # Fused AdamW on flat fp32 shards + multi-tensor sq-norm
 # --------------------------------------------------------------------------
 def fused_adamw(p, g, m, v, step, lr, beta1, beta2, eps, weight_decay):

@@ -55,6 +55,17 @@ def linear_cross_entropy(x, weight, labels, ignore_index=-100):
                            ignore_index=ignore_index)
 
 
+def causal_conv1d(x, weight, bias):
+    """Depthwise causal conv + silu (mamba xBC conv). x (b, l, C),
+    weight (C, W), bias (C)."""
+    b, l, C = x.shape
+    W = weight.shape[1]
+    xt = x.transpose(1, 2).float()
+    y = F.conv1d(F.pad(xt, (W - 1, 0)), weight.float().unsqueeze(1),
+                 bias.float(), groups=C)
+    return F.silu(y).transpose(1, 2).to(x.dtype)
+
+
 def adamw_step(p, g, m, v, step, lr, beta1, beta2, eps, weight_decay):
     """In-place fp32 AdamW on flat tensors (the shard update)."""
     p.mul_(1 - lr * weight_decay)
