@@ -120,7 +120,7 @@ class Attention(nn.Module):
         self.qkv = nn.Linear(cfg.emb_dim, qkv_out, bias=False)
         self.proj = nn.Linear(cfg.emb_dim, cfg.emb_dim, bias=False)
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, cache=None):
         b, s, _ = x.shape
         qkv = self.qkv(x)
         q, k, v = qkv.split(
@@ -131,7 +131,20 @@ class Attention(nn.Module):
         k = k.view(b, s, self.kvheads, self.head_dim)
         v = v.view(b, s, self.kvheads, self.head_dim)
         q, k = ops.rope_apply(q, k, cos, sin)
-        o = ops.attention_causal(q, k, v)          # (b, s, nheads, head_dim)
+        if cache is not None:
+            # decode path with KV cache (speculator stage-2 generation);
+            # memory-bound small-batch attention runs through torch SDPA.
+            import torch.nn.functional as F
+            if cache.get("k") is not None:
+                k = torch.cat([cache["k"], k], dim=1)
+                v = torch.cat([cache["v"], v], dim=1)
+            cache["k"], cache["v"] = k, v
+            o = F.scaled_dot_product_attention(
+                q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+                is_causal=(s == k.shape[1]),
+                enable_gqa=(self.kvheads != self.nheads)).transpose(1, 2)
+        else:
+            o = ops.attention_causal(q, k, v)      # (b, s, nheads, head_dim)
         return self.proj(o.reshape(b, s, -1))
 
     def reset_parameters(self):
@@ -167,14 +180,15 @@ class LlamaBlock(nn.Module):
         self.mlp_norm = RMSNorm(cfg.emb_dim, cfg.norm_eps)
         self.mlp = SwiGLU(cfg)
 
-    def forward(self, x, cos, sin):
-        if getattr(self, "_ac_enabled", False) and torch.is_grad_enabled():
+    def forward(self, x, cos, sin, cache=None):
+        if cache is None and getattr(self, "_ac_enabled", False) \
+                and torch.is_grad_enabled():
             return torch.utils.checkpoint.checkpoint(
                 self._forward_impl, x, cos, sin, use_reentrant=False)
-        return self._forward_impl(x, cos, sin)
+        return self._forward_impl(x, cos, sin, cache)
 
-    def _forward_impl(self, x, cos, sin):
-        x = x + self.attn(self.attn_norm(x), cos, sin)
+    def _forward_impl(self, x, cos, sin, cache=None):
+        x = x + self.attn(self.attn_norm(x), cos, sin, cache)
         x = x + self.mlp(self.mlp_norm(x))
         return x
 
@@ -193,10 +207,12 @@ class Llama(nn.Module):
         self.norm = RMSNorm(cfg.emb_dim, cfg.norm_eps)
         self.lm_head = nn.Linear(cfg.emb_dim, cfg.src_vocab_size, bias=False)
 
-    def forward(self, tokens, labels=None):
+    def forward(self, tokens, labels=None, include_embeds=False):
         """tokens (b, s) int64 -> logits (b, s, V); with labels also the
         mean CE loss via the chunked fused kernel (never materializes the
-        fp32 softmax — SURVEY.md hard-part 6)."""
+        fp32 softmax — SURVEY.md hard-part 6). include_embeds additionally
+        returns the final hidden states (speculator training input,
+        reference analog: EmbedLLaMA, train_speculator_utils.py:430-466)."""
         b, s = tokens.shape
         x = self.embedding(tokens)
         cos, sin = self.rot_emb.get(s, x.device)
@@ -204,8 +220,45 @@ class Llama(nn.Module):
             x = layer(x, cos, sin)
         x = self.norm(x)
         if labels is not None:
-            return ops.linear_cross_entropy(x, self.lm_head.weight, labels)
-        return self.lm_head(x)
+            loss = ops.linear_cross_entropy(x, self.lm_head.weight, labels)
+            return (loss, x) if include_embeds else loss
+        logits = self.lm_head(x)
+        return (logits, x) if include_embeds else logits
+
+    @torch.no_grad()
+    def generate(self, input_ids, max_new_tokens, temperature=1.0,
+                 do_sample=True, include_embeds=False):
+        """KV-cached autoregressive generation, optionally returning the
+        hidden state of each generated position (reference analog:
+        speculator/train_speculator_utils.py:28-118 generate())."""
+        b, s0 = input_ids.shape
+        caches = [{} for _ in self.layers]
+        tokens = input_ids
+        embeds = []
+        cur = input_ids
+        pos = 0
+        for step in range(max_new_tokens):
+            x = self.embedding(cur)
+            total = pos + cur.shape[1]
+            cos, sin = self.rot_emb.get(total, x.device)
+            cos_c, sin_c = cos[pos:total], sin[pos:total]
+            for layer, cache in zip(self.layers, caches):
+                x = layer(x, cos_c, sin_c, cache)
+            x = self.norm(x)
+            h_last = x[:, -1:]
+            logits = self.lm_head(h_last)[:, -1]
+            if do_sample:
+                probs = torch.softmax(logits.float() / temperature, dim=-1)
+                nxt = torch.multinomial(probs, 1)
+            else:
+                nxt = logits.argmax(-1, keepdim=True)
+            embeds.append(h_last)
+            tokens = torch.cat([tokens, nxt], dim=1)
+            pos = total
+            cur = nxt
+        if include_embeds:
+            return tokens, torch.cat(embeds, dim=1)
+        return tokens
 
     def reset_parameters(self):
         nn.init.trunc_normal_(self.embedding.weight, mean=0.0, std=0.02)

@@ -435,7 +435,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   const int kvh = h / (H / KVH);
   const int ngrp = H / KVH;
   const int kv0 = blockIdx.x * 128 + wid * KVB;  // this wave's keys
-  const int my_key = kv0 + col;
 
   const long long qrow_stride = (long long)H * D;
   const long long krow_stride = (long long)KVH * D;
@@ -596,23 +595,25 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
     __syncthreads();
   }
 
-  // write dK/dV: GQA head-groups collide on (b, key, kvh) -> atomicAdd
-  // fp32 when ngrp>1, plain add otherwise (buffers are zero-initialized).
-  float* dkp = dkg + ((long long)b * S * KVH + (long long)kvh) * D +
-               (long long)my_key * krow_stride;
-  float* dvp = dvg + ((long long)b * S * KVH + (long long)kvh) * D +
-               (long long)my_key * krow_stride;
+  // write dK/dV. D-layout: row i = key_rel = DROW(r,hb), col j = feature
+  // = t*32 + (lane&31). GQA head-groups collide on (b, key, kvh) ->
+  // atomicAdd fp32 when ngrp>1, plain add otherwise (buffers zeroed).
 #pragma unroll
   for (int t = 0; t < NT; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      const int dx = t * 32 + DROW(r, hb);
+      const int key_abs = kv0 + DROW(r, hb);
+      const int feat = t * 32 + col;
+      float* dkp = dkg + ((long long)b * S + key_abs) * (long long)(KVH * D) +
+                   (long long)kvh * D;
+      float* dvp = dvg + ((long long)b * S + key_abs) * (long long)(KVH * D) +
+                   (long long)kvh * D;
       if (ngrp > 1) {
-        atomicAdd(&dkp[dx], accDK[t][r]);
-        atomicAdd(&dvp[dx], accDV[t][r]);
+        atomicAdd(&dkp[feat], accDK[t][r]);
+        atomicAdd(&dvp[feat], accDV[t][r]);
       } else {
-        dkp[dx] += accDK[t][r];
-        dvp[dx] += accDV[t][r];
+        dkp[feat] += accDK[t][r];
+        dvp[feat] += accDV[t][r];
       }
     }
   }

@@ -3,12 +3,66 @@ torch intermediates computed from the SAME bf16 inputs and the kernel's
 own lse, and prints the error structure (per q-block / dk-block) to
 localize fragment-layout bugs. Run: python tests/debug_attn_bwd.py"""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
 import torch
 
 from fms_fsdp_amd import _C
 
+import itertools
+
 torch.manual_seed(0)
 dev = "cuda:0"
+
+SHAPES = [(1, 128, 1, 1, 128), (1, 128, 2, 2, 128), (2, 256, 4, 2, 128)]
+
+
+def run_shape(b, s, h, kvh, d):
+    print(f"######## shape b{b} s{s} h{h} kvh{kvh} d{d}")
+    scale = d ** -0.5
+    q = torch.randn(b, s, h, d, device=dev, dtype=torch.bfloat16)
+    k = torch.randn(b, s, kvh, d, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(b, s, kvh, d, device=dev, dtype=torch.bfloat16)
+    do = torch.randn(b, s, h, d, device=dev, dtype=torch.bfloat16)
+    o, lse = _C.attn_fwd(q, k, v)
+    dq, dk, dv = _C.attn_bwd(do, q, k, v, o, lse)
+    # reference per (b, h)
+    rep = h // kvh
+    dq_ref = torch.zeros_like(dq, dtype=torch.float32)
+    dk_ref = torch.zeros(b, s, kvh, d, device=dev)
+    dv_ref = torch.zeros(b, s, kvh, d, device=dev)
+    mask = torch.ones(s, s, device=dev, dtype=torch.bool).triu(1)
+    for bi in range(b):
+        for hi in range(h):
+            Q = q.float()[bi, :, hi]
+            K = k.float()[bi, :, hi // rep]
+            V = v.float()[bi, :, hi // rep]
+            dO = do.float()[bi, :, hi]
+            O = o.float()[bi, :, hi]
+            L = lse.float()[bi, hi]
+            S_ = Q @ K.t() * scale
+            P = torch.exp(S_ - L[:, None]).masked_fill(mask, 0.0)
+            dP = dO @ V.t()
+            delta = (dO * O).sum(-1)
+            dS = P * (dP - delta[:, None]) * scale
+            dq_ref[bi, :, hi] = dS @ K
+            dk_ref[bi, :, hi // rep] += dS.t() @ Q
+            dv_ref[bi, :, hi // rep] += P.t() @ dO
+    for name, got, ref in [("dq", dq.float(), dq_ref),
+                           ("dk", dk.float(), dk_ref),
+                           ("dv", dv.float(), dv_ref)]:
+        err = (got - ref).abs().max().item()
+        rel = err / ref.abs().max().clamp(min=1e-6).item()
+        print(f"  {name}: max abs {err:.3e} rel {rel:.3e} "
+              f"{'OK' if rel < 0.05 else 'FAIL'}")
+
+
+for shp in SHAPES:
+    run_shape(*shp)
+
 b, s, h, kvh, d = 1, 128, 1, 1, 128
 scale = d ** -0.5
 
