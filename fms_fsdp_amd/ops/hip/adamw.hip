@@ -12,9 +12,11 @@ __global__ void adamw_kernel(float* __restrict__ p,
                              float* __restrict__ m,
                              float* __restrict__ v,
                              long long n4, float lr, float b1, float b2,
-                             float eps, float wd, float bc1, float bc2) {
+                             float eps, float wd, float bc1, float bc2,
+                             const float* __restrict__ gscale) {
   const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n4) return;
+  const float gs = gscale ? *gscale : 1.f;
   f32x4 pv = ((f32x4*)p)[i];
   f32x4 mv = ((f32x4*)m)[i];
   f32x4 vv = ((f32x4*)v)[i];
@@ -22,11 +24,11 @@ __global__ void adamw_kernel(float* __restrict__ p,
   if (GRAD_BF16) {
     const bf16x4 g = ((const bf16x4*)g_)[i];
 #pragma unroll
-    for (int j = 0; j < 4; ++j) gf[j] = bf2f(g.v[j]);
+    for (int j = 0; j < 4; ++j) gf[j] = bf2f(g.v[j]) * gs;
   } else {
     const f32x4 g = ((const f32x4*)g_)[i];
 #pragma unroll
-    for (int j = 0; j < 4; ++j) gf[j] = g.v[j];
+    for (int j = 0; j < 4; ++j) gf[j] = g.v[j] * gs;
   }
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
@@ -68,17 +70,19 @@ extern "C" {
 void launch_adamw(float* p, const void* g, int grad_is_bf16, float* m,
                   float* v, long long n, float lr, float b1, float b2,
                   float eps, float wd, float bc1, float bc2,
-                  hipStream_t stream) {
+                  const float* gscale, hipStream_t stream) {
   // flat shards are 128-element aligned; n % 4 == 0 guaranteed
   const long long n4 = n / 4;
   const int block = 256;
   const long long grid = (n4 + block - 1) / block;
   if (grad_is_bf16)
     adamw_kernel<true><<<(int)grid, block, 0, stream>>>(p, g, m, v, n4, lr, b1,
-                                                        b2, eps, wd, bc1, bc2);
+                                                        b2, eps, wd, bc1, bc2,
+                                                        gscale);
   else
     adamw_kernel<false><<<(int)grid, block, 0, stream>>>(p, g, m, v, n4, lr, b1,
-                                                         b2, eps, wd, bc1, bc2);
+                                                         b2, eps, wd, bc1, bc2,
+                                                         gscale);
 }
 
 void launch_sqnorm(const void* t, int is_bf16, float* out, long long n,

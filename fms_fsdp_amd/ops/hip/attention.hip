@@ -56,7 +56,7 @@ __device__ __forceinline__ bf16x8v pack_pT_chunk(const float* p) {
 // grid.x = s/128, grid.y = b*h.
 // ---------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256) void attn_fwd_kernel(
+__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const short* __restrict__ qg, const short* __restrict__ kg,
     const short* __restrict__ vg, short* __restrict__ og,
     float* __restrict__ lseg, int B, int S, int H, int KVH, float scale) {
@@ -131,69 +131,70 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- S^T = K @ Q^T : two 32-key sub-tiles ----
-    f32x16 accS[2];
-    accS[0] = (f32x16)(0.f);
-    accS[1] = (f32x16)(0.f);
+    // ---- two 32-key sub-tiles, each with its own online-softmax pass.
+    // Register economy: one live accS/p set (16 regs) instead of two,
+    // keeping total VGPR+AGPR under 256 for 2 waves/SIMD occupancy.
+    // defer-max (guide T13, THR=8): the O/l rescale runs only when the
+    // sub-tile max exceeds the running max by more than THR; P is then
+    // bounded by e^THR which the fp32 accumulate tolerates. Decision is
+    // made BEFORE this sub-tile's P is exponentiated (the safe order).
 #pragma unroll
     for (int kt = 0; kt < 2; ++kt) {
+      const int kv32 = kv0 + kt * 32;
+      if (kv32 > qw + 31) break;      // wave-uniform causal tile skip
+      f32x16 accS = (f32x16)(0.f);
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         const int row = kt * 32 + col;
         const int inrow = c * 32 + hb * 16;
         const bf16x8v a =
             *(const bf16x8v*)(k_lds + ((row * (D * 2) + inrow) ^ ((row & KSWZ) << 4)));
-        accS[kt] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qb[c], accS[kt], 0, 0, 0);
+        accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qb[c], accS, 0, 0, 0);
       }
-    }
-
-    // ---- online softmax (lane-local q-row) ----
-    float p[32];
-    const bool partial = (kv0 + KVB) > (qw + 1);  // any masking possible
-    float mt = -1e30f;
-#pragma unroll
-    for (int kt = 0; kt < 2; ++kt)
+      float p[16];
+      const bool partial = (kv32 + 32) > (qw + 1);
+      float mt = -1e30f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        float sv = accS[kt][r] * scale;
-        if (partial) {
-          const int key = kv0 + kt * 32 + DROW(r, hb);
-          if (key > my_q) sv = -1e30f;
-        }
-        p[kt * 16 + r] = sv;
+        float sv = accS[r] * scale;
+        if (partial && (kv32 + DROW(r, hb)) > my_q) sv = -1e30f;
+        p[r] = sv;
         mt = fmaxf(mt, sv);
       }
-    mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
-    const float m_new = fmaxf(m_run, mt);
-    const float alpha = __expf(m_run - m_new);
-    float s_own = 0.f;
+      mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
+      float alpha = 1.f;
+      if (mt > m_run + 8.f) {          // defer-max threshold
+        alpha = __expf(m_run - mt);
+        m_run = mt;
 #pragma unroll
-    for (int i = 0; i < 32; ++i) {
-      p[i] = __expf(p[i] - m_new);
-      s_own += p[i];
-    }
-    l_run = l_run * alpha + s_own + __shfl_xor(s_own, 32, 64);
-    m_run = m_new;
+        for (int t = 0; t < NT; ++t)
 #pragma unroll
-    for (int t = 0; t < NT; ++t)
+          for (int r = 0; r < 16; ++r) accO[t][r] *= alpha;
+      }
+      float s_own = 0.f;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) accO[t][r] *= alpha;
+      for (int r = 0; r < 16; ++r) {
+        p[r] = __expf(p[r] - m_run);
+        s_own += p[r];
+      }
+      l_run = l_run * alpha + s_own + __shfl_xor(s_own, 32, 64);
 
-    // ---- PV: O^T += V^T @ P^T ----
-    bf16x8v pb[4];
+      // PV for this sub-tile: keys kv32..kv32+31 = chunks 2kt, 2kt+1
+      const bf16x8v pb0 = pack_pT_chunk(p);
+      const bf16x8v pb1 = pack_pT_chunk(p + 8);
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc) pb[kc] = pack_pT_chunk(p + kc * 8);
-#pragma unroll
-    for (int t = 0; t < NT; ++t) {
-#pragma unroll
-      for (int kc = 0; kc < 4; ++kc) {
+      for (int t = 0; t < NT; ++t) {
         const int row = t * 32 + col;   // dv row
-        const int inrow = kc * 32 + hb * 16;
-        const bf16x8v a = *(const bf16x8v*)(
-            vt_lds + ((row * (KVB * 2) + inrow) ^ ((row & 7) << 4)));
+        const int inrow0 = (kt * 2) * 32 + hb * 16;
+        const bf16x8v a0 = *(const bf16x8v*)(
+            vt_lds + ((row * (KVB * 2) + inrow0) ^ ((row & 7) << 4)));
         accO[t] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, pb[kc], accO[t], 0, 0, 0);
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, pb0, accO[t], 0, 0, 0);
+        const int inrow1 = (kt * 2 + 1) * 32 + hb * 16;
+        const bf16x8v a1 = *(const bf16x8v*)(
+            vt_lds + ((row * (KVB * 2) + inrow1) ^ ((row & 7) << 4)));
+        accO[t] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, pb1, accO[t], 0, 0, 0);
       }
     }
     __syncthreads();
@@ -406,7 +407,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 //   dK[key][dk] = sum_q dS^T[key][q] Q[q][dk]   (A=dS from p_lds, B=Qt)
 // ---------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const short* __restrict__ dog, const short* __restrict__ qg,
     const short* __restrict__ kg, const short* __restrict__ vg,
     const float* __restrict__ lseg, const float* __restrict__ deltag,
@@ -420,9 +421,10 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   // per-wave carves
   char* k_lds = smem;                                  // [4][KVB][D*2]
   char* v_lds = smem + 4 * KVB * D * 2;                // [4][KVB][D*2]
-  char* qt_lds = smem + 8 * KVB * D * 2;               // [D][32*2] shared
-  char* dot_lds = qt_lds + D * 64;                     // [D][32*2] shared
-  char* p_lds = dot_lds + D * 64;                      // [4][KVB][32*2]
+  // ONE shared transpose buffer, staged with dO^T then re-staged with Q^T
+  // each q-iteration: keeps total LDS at 80 KB -> 2 blocks/CU.
+  char* t_lds = smem + 8 * KVB * D * 2;                // [D][32*2] shared
+  char* p_lds = t_lds + D * 64;                        // [4][KVB][32*2]
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -474,19 +476,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   // the same q range (block-uniform barriers), masking handles the rest.
   const int q_start = (blockIdx.x * 128) / 32 * 32;
   for (int q0 = q_start; q0 < S; q0 += 32) {
-    // stage Qt / dOt (transposed, shared): thread t: q = t&31, dk grp t>>5
+    // stage dO^T (transposed, shared): thread t: q = t&31, dk grp t>>5
     {
       const int t256 = threadIdx.x;
       const int q = t256 & 31;
       const int ng = 256 / 32;                  // 8 groups over dk
       const int dk0 = (t256 >> 5) * (D / ng);
-      const short* qp = qbase + (long long)(q0 + q) * qrow_stride + dk0;
       const short* dp = dobase + (long long)(q0 + q) * qrow_stride + dk0;
 #pragma unroll
       for (int j = 0; j < D / ng; ++j) {
         const int dk = dk0 + j;
-        *(short*)(qt_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = qp[j];
-        *(short*)(dot_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = dp[j];
+        *(short*)(t_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = dp[j];
       }
     }
     __syncthreads();
@@ -536,7 +536,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
       const int key = DROW(r, hb);
       *(short*)(my_p + ((key * 64 + col * 2) ^ ((key & 3) << 4))) = f2bf(pv[r]);
     }
-    __syncthreads();
+    // (p_lds is wave-private: lgkmcnt ordering suffices, no barrier)
     // dV[key][dv] += P(A) @ dOt(B): A[i=key][k=q] from p_lds
     {
       bf16x8v pa[2];
@@ -557,14 +557,26 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
           const int row = t * 32 + col;
           const int inrow = kc * 32 + hb * 16;
           const bf16x8v bb = *(const bf16x8v*)(
-              dot_lds + ((row * 64 + inrow) ^ ((row & 3) << 4)));
+              t_lds + ((row * 64 + inrow) ^ ((row & 3) << 4)));
           accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kc], bb,
                                                              accDV[t], 0, 0, 0);
         }
       }
     }
-    __syncthreads();
-    // overwrite p_lds with dS^T, then dK
+    __syncthreads();  // every wave done with the dO image
+    // re-stage t_lds with Q^T; overwrite p_lds with dS^T
+    {
+      const int t256 = threadIdx.x;
+      const int q = t256 & 31;
+      const int ng = 256 / 32;
+      const int dk0 = (t256 >> 5) * (D / ng);
+      const short* qp = qbase + (long long)(q0 + q) * qrow_stride + dk0;
+#pragma unroll
+      for (int j = 0; j < D / ng; ++j) {
+        const int dk = dk0 + j;
+        *(short*)(t_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = qp[j];
+      }
+    }
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key = DROW(r, hb);
@@ -586,7 +598,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
           const int row = t * 32 + col;
           const int inrow = kc * 32 + hb * 16;
           const bf16x8v bb = *(const bf16x8v*)(
-              qt_lds + ((row * 64 + inrow) ^ ((row & 3) << 4)));
+              t_lds + ((row * 64 + inrow) ^ ((row & 3) << 4)));
           accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[kc], bb,
                                                              accDK[t], 0, 0, 0);
         }
@@ -650,7 +662,7 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
     attn_bwd_dq_kernel<128><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
-    const int lds_dkv = 8 * 32 * 128 * 2 + 2 * 128 * 64 + 4 * 32 * 64;
+    const int lds_dkv = 8 * 32 * 128 * 2 + 128 * 64 + 4 * 32 * 64;
     attn_bwd_dkv_kernel<128><<<grid, 256, lds_dkv, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (float*)dk, (float*)dv, B, S, H, KVH, scale);
@@ -659,7 +671,7 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
     attn_bwd_dq_kernel<64><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
-    const int lds_dkv = 8 * 32 * 64 * 2 + 2 * 64 * 64 + 4 * 32 * 64;
+    const int lds_dkv = 8 * 32 * 64 * 2 + 64 * 64 + 4 * 32 * 64;
     attn_bwd_dkv_kernel<64><<<grid, 256, lds_dkv, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (float*)dk, (float*)dv, B, S, H, KVH, scale);

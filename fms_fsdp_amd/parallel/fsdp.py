@@ -108,10 +108,18 @@ class FlatUnit:
             if self.S == 1:
                 self.param_shard = self.flat_param  # same storage
             self.master_shard = self.param_shard.float()
-            self.grad_shard = torch.zeros_like(self.master_shard)
             self.exp_avg = torch.zeros_like(self.master_shard)
             self.exp_avg_sq = torch.zeros_like(self.master_shard)
             self.flat_grad = torch.zeros(self.total, dtype=self.param_dtype, device=dev)
+            # grads stay in the param/reduce dtype end-to-end (bf16 reduce
+            # like the reference's bfSixteen policy); for S==1 the "shard"
+            # aliases flat_grad so the optimizer reads autograd's buffer
+            # with zero extra memory traffic.
+            if self.S > 1:
+                self.grad_shard = torch.zeros(self.shard_size,
+                                              dtype=self.param_dtype, device=dev)
+            else:
+                self.grad_shard = self.flat_grad
             self._point_params_into_flat()
             self._set_grad_views()
             self._gathered = True
@@ -185,27 +193,12 @@ class FlatUnit:
         ctx = torch.cuda.stream(rs_stream) if rs_stream is not None else nullcontext()
         with ctx:
             if self.S > 1:
-                rs_out = torch.empty(self.shard_size, dtype=self.flat_grad.dtype,
-                                     device=self.device)
-                dist.reduce_scatter_tensor(rs_out, self.flat_grad,
+                dist.reduce_scatter_tensor(self.grad_shard, self.flat_grad,
                                            op=dist.ReduceOp.SUM, group=self.shard_group)
-                rsf = rs_out.float()
-                if total_dp > 1:
-                    rsf /= total_dp
-                if self.replicate_group is not None:
-                    dist.all_reduce(rsf, group=self.replicate_group)
-                self.grad_shard.add_(rsf)
-                del rs_out
-            else:
-                rsf = self.flat_grad.float()
-                if total_dp > 1:
-                    rsf /= total_dp
-                if self.replicate_group is not None:
-                    dist.all_reduce(rsf, group=self.replicate_group)
-                self.grad_shard.add_(rsf)
-            # grads for this unit are consumed; zero the bf16 buffer for the
-            # next backward (accumulation now lives in fp32 grad_shard)
-            self.flat_grad.zero_()
+            if self.replicate_group is not None:
+                dist.all_reduce(self.grad_shard, group=self.replicate_group)
+            if total_dp > 1:
+                self.grad_shard.div_(total_dp)
             if rs_stream is not None:
                 self._grad_event = torch.cuda.Event()
                 self._grad_event.record(rs_stream)
@@ -373,29 +366,25 @@ class ShardedModel(nn.Module):
     # ---------------- training utilities ----------------
 
     def clip_grad_norm_(self, max_norm):
-        """Global grad-norm over fp32 grad shards + clip (reference:
-        train_utils.py:96 model.clip_grad_norm_)."""
+        """Global grad-norm over the bf16 grad shards (reference:
+        train_utils.py:96 model.clip_grad_norm_). The clip factor is NOT
+        applied as an extra pass over the grads — it is folded into the
+        fused AdamW kernel (one less 2x-shard-size memory sweep)."""
         for u in self.all_units:
             u.wait_grads()
         local = ops.sq_norm([u.grad_shard for u in self.all_units])
         if self.shard_group is not None:
             dist.all_reduce(local, group=self.shard_group)
         total_norm = local.sqrt()
-        clip = max_norm / (total_norm + 1e-6)
-        if self.is_cuda:
-            coef = torch.clamp(clip, max=1.0)
-            for u in self.all_units:
-                u.grad_shard.mul_(coef)
-        else:
-            c = min(clip.item(), 1.0)
-            if c < 1.0:
-                for u in self.all_units:
-                    u.grad_shard.mul_(c)
+        self._clip_coef = torch.clamp(max_norm / (total_norm + 1e-6), max=1.0)
         return total_norm
 
     def zero_grad(self, set_to_none=False):
+        self._clip_coef = None
         for u in self.all_units:
-            u.grad_shard.zero_()
+            u.flat_grad.zero_()
+            if u.S > 1:
+                u.grad_shard.zero_()
             if not u._grads_ready_views:
                 u._set_grad_views()
 
@@ -423,11 +412,12 @@ class ShardedAdamW:
         self.step_count += 1
         lr = self.param_groups[0]["lr"]
         b1, b2 = self.betas
+        gscale = getattr(self.m, "_clip_coef", None)
         for u in self.m.all_units:
             u.wait_grads()
             ops.fused_adamw(u.master_shard, u.grad_shard, u.exp_avg,
                             u.exp_avg_sq, self.step_count, lr, b1, b2,
-                            self.eps, self.weight_decay)
+                            self.eps, self.weight_decay, grad_scale=gscale)
             u.publish_master_to_shard()
             u.mark_stale()
 
