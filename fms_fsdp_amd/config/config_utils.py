@@ -106,6 +106,27 @@ def get_model_config(model_variant):
             "pad_vocab_size_multiple": 16,
             "tie_embeddings": False,
         }
+    if model_variant == "mamba_test":
+        # tiny CPU-trivial hybrid used by the test suite / smoke runs
+        return {
+            "d_model": 64,
+            "d_intermediate": 128,
+            "n_layer": 2,
+            "vocab_size": 512,
+            "ssm_cfg": {"layer": "Mamba2"},
+            "attn_layer_idx": [1],
+            "attn_cfg": {"causal": True, "d_conv": 0, "head_dim": 64,
+                         "num_heads": 1, "num_heads_kv": 1,
+                         "rotary_emb_dim": 16},
+            "rms_norm": True,
+            "residual_in_fp32": True,
+            "fused_add_norm": True,
+            "pad_vocab_size_multiple": 16,
+            "tie_embeddings": False,
+            "d_state": 16,
+            "headdim": 16,
+            "chunk_size": 64,
+        }
     if model_variant == "mamba_2.8b":
         return {
             "d_model": 2560,
