@@ -191,11 +191,30 @@ def linear_cross_entropy(x, weight, labels, ignore_index=-100):
 
 
 # --------------------------------------------------------------------------
-# Causal depthwise conv1d + silu (mamba path). Runs through torch (MIOpen
-# depthwise conv) on GPU for now; a dedicated fused HIP kernel is on the
-# worklist (SURVEY.md §2.3 causal-conv1d row).
+# Fused causal depthwise conv1d + silu (mamba xBC conv)
 # --------------------------------------------------------------------------
+class _CausalConv1dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ext = _require_ext("causal_conv1d")
+        x = x.contiguous()
+        wb = weight.contiguous().bfloat16()
+        bf = bias.float().contiguous()
+        y = ext.cconv_fwd(x, wb, bf)
+        ctx.save_for_backward(x, wb, bf)
+        ctx.wdtype = weight.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, wb, bf = ctx.saved_tensors
+        dx, dw, db = _C.cconv_bwd(dy.contiguous(), x, wb, bf)
+        return dx, dw.to(ctx.wdtype), db.to(ctx.wdtype)
+
+
 def causal_conv1d(x, weight, bias):
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _CausalConv1dFn.apply(x, weight, bias)
     return reference.causal_conv1d(x, weight, bias)
 
 

@@ -29,6 +29,11 @@ void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
 void launch_attn_bwd(const void*, const void*, const void*, const void*,
                      const void*, const float*, void*, void*, void*, float*,
                      int, int, int, int, int, float, hipStream_t);
+void launch_cconv_fwd(const void*, const void*, const float*, void*, int,
+                      int, int, int, hipStream_t);
+void launch_cconv_bwd(const void*, const void*, const void*, const float*,
+                      void*, void*, float*, float*, int, int, int, int,
+                      hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -173,7 +178,39 @@ std::tuple<Tensor, Tensor, Tensor> attn_bwd(Tensor do_, Tensor q, Tensor k,
   return {dq, dk.to(torch::kBFloat16), dv.to(torch::kBFloat16)};
 }
 
+Tensor cconv_fwd(Tensor x, Tensor w, Tensor bias) {
+  CHECK_BF16_CONTIG(x);
+  CHECK_BF16_CONTIG(w);
+  const int b = x.size(0), l = x.size(1), c = x.size(2);
+  const int W = w.size(1);
+  TORCH_CHECK(c % 8 == 0 && W <= 4);
+  TORCH_CHECK(bias.scalar_type() == torch::kFloat32);
+  auto y = torch::empty_like(x);
+  launch_cconv_fwd(x.data_ptr(), w.data_ptr(), bias.data_ptr<float>(),
+                   y.data_ptr(), b * l, l, c, W, cur_stream());
+  return y;
+}
+
+std::tuple<Tensor, Tensor, Tensor> cconv_bwd(Tensor dy, Tensor x, Tensor w,
+                                             Tensor bias) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(x);
+  const int b = x.size(0), l = x.size(1), c = x.size(2);
+  const int W = w.size(1);
+  auto g = torch::empty_like(x);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({c, W}, x.options().dtype(torch::kFloat32));
+  auto db = torch::zeros({c}, x.options().dtype(torch::kFloat32));
+  launch_cconv_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                   bias.data_ptr<float>(), g.data_ptr(), dx.data_ptr(),
+                   dw.data_ptr<float>(), db.data_ptr<float>(), b * l, l, c, W,
+                   cur_stream());
+  return {dx, dw, db};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("cconv_fwd", &cconv_fwd);
+  mod.def("cconv_bwd", &cconv_bwd);
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);
   mod.def("rmsnorm_bwd", &rmsnorm_bwd);
   mod.def("rope_fwd", &rope_fwd);

@@ -199,3 +199,26 @@ def test_model_gpu_step():
         losses.append(loss.item())
     assert losses[-1] < losses[0], losses
     assert all(l == l for l in losses), losses
+
+
+def test_causal_conv1d():
+    torch.manual_seed(7)
+    from fms_fsdp_amd import _C
+    b, l, C, W = 2, 64, 256, 4
+    x = torch.randn(b, l, C, device=dev(), dtype=torch.bfloat16)
+    w = torch.randn(C, W, device=dev(), dtype=torch.bfloat16)
+    bias = torch.randn(C, device=dev())
+    y = _C.cconv_fwd(x, w, bias)
+    ref = reference.causal_conv1d(x, w.float(), bias)
+    assert relerr(y, ref) < 2e-2
+
+    xf = x.float().requires_grad_()
+    wf = w.float().requires_grad_()
+    bfp = bias.clone().requires_grad_()
+    yr = reference.causal_conv1d(xf, wf, bfp)
+    dy = torch.randn_like(yr)
+    yr.backward(dy)
+    dx, dw, db = _C.cconv_bwd(dy.bfloat16(), x, w, bias)
+    assert relerr(dx, xf.grad) < 5e-2
+    assert relerr(dw, wf.grad) < 5e-2
+    assert relerr(db, bfp.grad) < 5e-2
