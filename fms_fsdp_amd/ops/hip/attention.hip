@@ -118,15 +118,21 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
         *(f32x4*)(k_lds + ((row * BYTES_PER_ROW + cb) ^ ((row & KSWZ) << 4))) =
             *(const f32x4*)(kbase + g);
       }
-      // V transposed: thread t: key = t&63, dv block = (t>>6)*32
+      // V transposed: thread t: key = t&63, dv block = (t>>6)*32.
+      // Loads are vectorized 16B (guide G13: never scalar bf16 loads);
+      // the transposed LDS writes scatter but stay cheap vs the MFMAs.
       const int key = t256 & 63;
       const int dv0 = (t256 >> 6) * (D / 4);
       const short* vp = vbase + (long long)(kv0 + key) * krow_stride + dv0;
 #pragma unroll
-      for (int j = 0; j < D / 4; ++j) {
-        const int dv = dv0 + j;
-        *(short*)(vt_lds +
-                  ((dv * (KVB * 2) + key * 2) ^ ((dv & 7) << 4))) = vp[j];
+      for (int jj = 0; jj < D / 32; ++jj) {
+        const bf16x8v vv = *(const bf16x8v*)(vp + jj * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int dv = dv0 + jj * 8 + e;
+          *(__bf16*)(vt_lds +
+                     ((dv * (KVB * 2) + key * 2) ^ ((dv & 7) << 4))) = vv[e];
+        }
       }
     }
     __syncthreads();
@@ -331,10 +337,14 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       const int dk0 = (t256 / KVB) * (D / ng);
       const short* kp = kbase + (long long)(kv0 + key) * krow_stride + dk0;
 #pragma unroll
-      for (int j = 0; j < D / ng; ++j) {
-        const int dk = dk0 + j;
-        *(short*)(kt_lds + ((dk * (KVB * 2) + key * 2) ^ ((dk & 3) << 4))) =
-            kp[j];
+      for (int jj = 0; jj < D / ng / 8; ++jj) {
+        const bf16x8v kv8 = *(const bf16x8v*)(kp + jj * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int dk = dk0 + jj * 8 + e;
+          *(__bf16*)(kt_lds + ((dk * (KVB * 2) + key * 2) ^ ((dk & 3) << 4))) =
+              kv8[e];
+        }
       }
     }
     __syncthreads();
@@ -484,9 +494,13 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       const int dk0 = (t256 >> 5) * (D / ng);
       const short* dp = dobase + (long long)(q0 + q) * qrow_stride + dk0;
 #pragma unroll
-      for (int j = 0; j < D / ng; ++j) {
-        const int dk = dk0 + j;
-        *(short*)(t_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = dp[j];
+      for (int jj = 0; jj < D / ng / 8; ++jj) {
+        const bf16x8v dv8 = *(const bf16x8v*)(dp + jj * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int dk = dk0 + jj * 8 + e;
+          *(__bf16*)(t_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = dv8[e];
+        }
       }
     }
     __syncthreads();
@@ -572,9 +586,13 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       const int dk0 = (t256 >> 5) * (D / ng);
       const short* qp = qbase + (long long)(q0 + q) * qrow_stride + dk0;
 #pragma unroll
-      for (int j = 0; j < D / ng; ++j) {
-        const int dk = dk0 + j;
-        *(short*)(t_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = qp[j];
+      for (int jj = 0; jj < D / ng / 8; ++jj) {
+        const bf16x8v qv8 = *(const bf16x8v*)(qp + jj * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int dk = dk0 + jj * 8 + e;
+          *(__bf16*)(t_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = qv8[e];
+        }
       }
     }
 #pragma unroll
