@@ -39,6 +39,14 @@ def main():
     from fms_fsdp_amd.utils.train import setup_environ_flags
 
     setup_environ_flags()
+    # hipBLASLt algo table tuned on MI355X (committed in-tree); read-only
+    tune_csv = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "tunableop_results0.csv")
+    if os.path.exists(tune_csv) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
+        os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+        os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+        os.environ["PYTORCH_TUNABLEOP_FILENAME"] = \
+            tune_csv.replace("results0", "results%d")
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
