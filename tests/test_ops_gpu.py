@@ -216,9 +216,9 @@ def test_causal_conv1d():
     wf = w.float().requires_grad_()
     bfp = bias.clone().requires_grad_()
     yr = reference.causal_conv1d(xf, wf, bfp)
-    dy = torch.randn_like(yr)
+    dy = torch.randn(b, l, C, device=dev())  # contiguous (b,l,C)
     yr.backward(dy)
-    dx, dw, db = _C.cconv_bwd(dy.bfloat16(), x, w, bias)
+    dx, dw, db = _C.cconv_bwd(dy.bfloat16().contiguous(), x, w, bias)
     assert relerr(dx, xf.grad) < 5e-2
     assert relerr(dw, wf.grad) < 5e-2
     assert relerr(db, bfp.grad) < 5e-2
