@@ -94,3 +94,16 @@ def test_do_ckpt_file_poll(tmp_path):
     assert do_ckpt(d)
     do_ckpt(d, reset=True)
     assert not do_ckpt(d)
+
+
+def test_train_speculator_entry_smoke(tmp_path):
+    """Full speculator entry point: tiny base model, dummy data, 2 stage-1
+    steps single-process."""
+    from speculator import train_speculator as ts
+    ts.main(model_variant="llama2_125m", use_dummy_dataset=True,
+            batch_size=1, seq_length=128, num_steps=2, report_interval=1,
+            checkpoint_interval=100, mixed_precision=False,
+            n_speculator_heads=2, speculator_width=64,
+            stage2_start_step=10, model_path="/nonexistent",
+            ckpt_save_path=str(tmp_path), ckpt_load_path=str(tmp_path),
+            vocab_size=256, learning_rate=1e-4, sharding_strategy="fsdp")
