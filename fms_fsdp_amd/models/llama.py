@@ -188,9 +188,10 @@ class LlamaBlock(nn.Module):
         return self._forward_impl(x, cos, sin, cache)
 
     def _forward_impl(self, x, cos, sin, cache=None):
-        x = x + self.attn(self.attn_norm(x), cos, sin, cache)
-        x = x + self.mlp(self.mlp_norm(x))
-        return x
+        h = self.attn(self.attn_norm(x), cos, sin, cache)
+        # fused residual-add + norm: y2 = rmsnorm(x + h), s = x + h
+        y2, s = ops.add_rmsnorm(x, h, self.mlp_norm.weight, self.mlp_norm.eps)
+        return s + self.mlp(y2)
 
     def reset_parameters(self):
         for m in (self.attn_norm, self.attn, self.mlp_norm, self.mlp):

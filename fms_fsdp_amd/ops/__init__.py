@@ -47,7 +47,8 @@ class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x2d, weight, rinv = ctx.saved_tensors
-        dx, dw = _C.rmsnorm_bwd(dy.contiguous().view_as(x2d), x2d, weight, rinv)
+        dx, dw = _C.rmsnorm_bwd(dy.contiguous().view_as(x2d), x2d, weight,
+                                rinv, None)
         return dx.view(ctx.shape), dw.to(weight.dtype), None
 
 
@@ -55,6 +56,37 @@ def rmsnorm(x, weight, eps=1e-6):
     if x.is_cuda:
         return _RMSNormFn.apply(x, weight, eps)
     return reference.rmsnorm(x, weight, eps)
+
+
+class _AddRMSNormFn(torch.autograd.Function):
+    """Fused s = x + res; y = rmsnorm(s): one HBM pass instead of two.
+    Backward fuses the residual grad into the rmsnorm dx kernel."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, eps):
+        ext = _require_ext("add_rmsnorm")
+        x2d = x.contiguous().view(-1, x.shape[-1])
+        r2d = res.contiguous().view_as(x2d)
+        y, s, rinv = ext.add_rmsnorm_fwd(x2d, r2d, weight, eps)
+        ctx.save_for_backward(s, weight, rinv)
+        ctx.shape = x.shape
+        return y.view(x.shape), s.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, dy, ds):
+        s, weight, rinv = ctx.saved_tensors
+        dx, dw = _C.rmsnorm_bwd(dy.contiguous().view_as(s), s, weight, rinv,
+                                ds.contiguous().view_as(s))
+        dx = dx.view(ctx.shape)
+        return dx, dx, dw.to(weight.dtype), None
+
+
+def add_rmsnorm(x, res, weight, eps=1e-6):
+    """(rmsnorm(x + res), x + res)"""
+    if x.is_cuda:
+        return _AddRMSNormFn.apply(x, res, weight, eps)
+    s = (x.float() + res.float()).to(x.dtype)
+    return reference.rmsnorm(s, weight, eps), s
 
 
 # --------------------------------------------------------------------------

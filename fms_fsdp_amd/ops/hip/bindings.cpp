@@ -12,7 +12,9 @@ extern "C" {
 void launch_rmsnorm_fwd(const void*, const void*, void*, float*, int, int,
                         float, hipStream_t);
 void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
-                        void*, float*, int, int, hipStream_t);
+                        const void*, void*, float*, int, int, hipStream_t);
+void launch_add_rmsnorm_fwd(const void*, const void*, const void*, void*,
+                            void*, float*, int, int, float, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, int, int,
                  int, int, int, hipStream_t);
 void launch_swiglu_fwd(const void*, void*, long long, int, hipStream_t);
@@ -58,16 +60,34 @@ std::tuple<Tensor, Tensor> rmsnorm_fwd(Tensor x, Tensor w, double eps) {
 }
 
 std::tuple<Tensor, Tensor> rmsnorm_bwd(Tensor dy, Tensor x, Tensor w,
-                                       Tensor rinv) {
+                                       Tensor rinv,
+                                       c10::optional<Tensor> dextra) {
   CHECK_BF16_CONTIG(dy);
   CHECK_BF16_CONTIG(x);
   const int rows = x.size(0), H = x.size(1);
   auto dx = torch::empty_like(x);
   auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
   launch_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
-                     rinv.data_ptr<float>(), dx.data_ptr(),
-                     dw.data_ptr<float>(), rows, H, cur_stream());
+                     rinv.data_ptr<float>(),
+                     dextra.has_value() ? dextra->data_ptr() : nullptr,
+                     dx.data_ptr(), dw.data_ptr<float>(), rows, H,
+                     cur_stream());
   return {dx, dw};
+}
+
+std::tuple<Tensor, Tensor, Tensor> add_rmsnorm_fwd(Tensor x, Tensor res,
+                                                   Tensor w, double eps) {
+  CHECK_BF16_CONTIG(x);
+  CHECK_BF16_CONTIG(res);
+  const int rows = x.size(0), H = x.size(1);
+  auto y = torch::empty_like(x);
+  auto s_out = torch::empty_like(x);
+  auto rinv = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  launch_add_rmsnorm_fwd(x.data_ptr(), res.data_ptr(), w.data_ptr(),
+                         y.data_ptr(), s_out.data_ptr(),
+                         rinv.data_ptr<float>(), rows, H, (float)eps,
+                         cur_stream());
+  return {y, s_out, rinv};
 }
 
 std::tuple<Tensor, Tensor> rope_fwd(Tensor q, Tensor k, Tensor cos,
@@ -213,6 +233,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("cconv_bwd", &cconv_bwd);
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);
   mod.def("rmsnorm_bwd", &rmsnorm_bwd);
+  mod.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
   mod.def("rope_fwd", &rope_fwd);
   mod.def("swiglu_fwd", &swiglu_fwd);
   mod.def("swiglu_bwd", &swiglu_bwd);

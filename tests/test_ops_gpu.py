@@ -34,7 +34,7 @@ def test_rmsnorm_fwd_bwd(rows, H):
     yf = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) * wf
     dy = torch.randn_like(yf)
     yf.backward(dy)
-    dx, dw = _C.rmsnorm_bwd(dy.bfloat16(), x, w, rinv)
+    dx, dw = _C.rmsnorm_bwd(dy.bfloat16(), x, w, rinv, None)
     assert relerr(dx, xf.grad) < 5e-2
     assert relerr(dw, wf.grad) < 5e-2
 
@@ -222,3 +222,24 @@ def test_causal_conv1d():
     assert relerr(dx, xf.grad) < 5e-2
     assert relerr(dw, wf.grad) < 5e-2
     assert relerr(db, bfp.grad) < 5e-2
+
+
+def test_add_rmsnorm():
+    torch.manual_seed(8)
+    from fms_fsdp_amd import _C
+    rows, H = 256, 4096
+    x = torch.randn(rows, H, device=dev(), dtype=torch.bfloat16)
+    r = torch.randn(rows, H, device=dev(), dtype=torch.bfloat16)
+    w = torch.randn(H, device=dev(), dtype=torch.bfloat16)
+    y, s, rinv = _C.add_rmsnorm_fwd(x, r, w, 1e-6)
+    sref = (x.float() + r.float()).bfloat16()
+    yref = reference.rmsnorm(sref, w, 1e-6)
+    assert relerr(s, sref) < 1e-2
+    assert relerr(y, yref) < 2e-2
+    # fused-dextra backward == plain backward + add
+    dy = torch.randn(rows, H, device=dev(), dtype=torch.bfloat16)
+    ds = torch.randn(rows, H, device=dev(), dtype=torch.bfloat16)
+    dx1, dw1 = _C.rmsnorm_bwd(dy, s, w, rinv, ds)
+    dx0, dw0 = _C.rmsnorm_bwd(dy, s, w, rinv, None)
+    assert relerr(dx1, (dx0.float() + ds.float())) < 2e-2
+    assert relerr(dw1, dw0) < 1e-4
