@@ -106,6 +106,20 @@ def get_model_config(model_variant):
             "pad_vocab_size_multiple": 16,
             "tie_embeddings": False,
         }
+    if model_variant in ("gpt_bigcode_starcoder", "gpt_bigcode_test"):
+        from fms_fsdp_amd.models.gpt_bigcode import GPTBigCodeConfig
+        if model_variant == "gpt_bigcode_test":
+            return GPTBigCodeConfig(src_vocab_size=256, emb_dim=128, nheads=1,
+                                    nlayers=2, max_expected_seq_len=256)
+        return GPTBigCodeConfig()
+    if model_variant in ("mixtral_8x7b", "mixtral_test"):
+        from fms_fsdp_amd.models.mixtral import MixtralConfig
+        if model_variant == "mixtral_test":
+            return MixtralConfig(src_vocab_size=256, emb_dim=64, nheads=1,
+                                 kvheads=1, nlayers=2, hidden_dim=128,
+                                 num_experts=4, top_k=2,
+                                 max_expected_seq_len=256)
+        return MixtralConfig()
     if model_variant == "mamba_test":
         # tiny CPU-trivial hybrid used by the test suite / smoke runs
         return {

@@ -49,9 +49,18 @@ def main(**kwargs):
     device = torch.device("cuda", local_rank) if torch.cuda.is_available() \
         else torch.device("cpu")
 
-    # base model
+    # base model (reference supports llama / gpt_bigcode / mixtral bases,
+    # train_speculator_utils.py:430-523 there)
     mcfg = get_model_config(cfg.model_variant)
-    model = Llama(mcfg)
+    arch = cfg.model_arch.lower()
+    if arch in ("embedgptbigcode", "gpt_bigcode"):
+        from fms_fsdp_amd.models.gpt_bigcode import GPTBigCode
+        model = GPTBigCode(mcfg)
+    elif arch in ("embedmixtral", "mixtral"):
+        from fms_fsdp_amd.models.mixtral import Mixtral
+        model = Mixtral(mcfg)
+    else:
+        model = Llama(mcfg)
     model.reset_parameters()
     if cfg.model_path and os.path.exists(cfg.model_path):
         if os.path.isfile(cfg.model_path):
@@ -70,6 +79,12 @@ def main(**kwargs):
 
     # parallel layout for the frozen base model
     tp_group = None
+    if cfg.sharding_strategy == "tp" and world_size > 1 and arch not in (
+            "embedllama", "llama"):
+        if rank == 0:
+            print("WARNING: TP base-model sharding implemented for llama; "
+                  f"running {arch} replicated")
+        cfg.sharding_strategy = "fsdp"
     if cfg.sharding_strategy == "tp" and world_size > 1:
         tp_size = min(cfg.tp_size, world_size)
         n_dp = world_size // tp_size

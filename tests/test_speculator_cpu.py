@@ -107,3 +107,48 @@ def test_train_speculator_entry_smoke(tmp_path):
             stage2_start_step=10, model_path="/nonexistent",
             ckpt_save_path=str(tmp_path), ckpt_load_path=str(tmp_path),
             vocab_size=256, learning_rate=1e-4, sharding_strategy="fsdp")
+
+
+def test_gpt_bigcode_base():
+    from fms_fsdp_amd.config import get_model_config
+    from fms_fsdp_amd.models.gpt_bigcode import GPTBigCode
+    torch.manual_seed(3)
+    m = GPTBigCode(get_model_config("gpt_bigcode_test"))
+    m.reset_parameters()
+    m.eval()
+    x = torch.randint(0, 256, (2, 16))
+    logits, emb = m(x, include_embeds=True)
+    assert logits.shape == (2, 16, 256) and emb.shape == (2, 16, 128)
+    toks = m.generate(x, 3, do_sample=False)
+    assert toks.shape == (2, 19)
+    # greedy cached generation == argmax of full forward
+    assert torch.equal(toks[:, 16], m(x)[:, -1].argmax(-1))
+
+
+def test_mixtral_base():
+    from fms_fsdp_amd.config import get_model_config
+    from fms_fsdp_amd.models.mixtral import Mixtral
+    torch.manual_seed(4)
+    m = Mixtral(get_model_config("mixtral_test"))
+    m.reset_parameters()
+    m.eval()
+    x = torch.randint(0, 256, (2, 16))
+    logits, emb = m(x, include_embeds=True)
+    assert logits.shape == (2, 16, 256) and emb.shape == (2, 16, 64)
+    toks = m.generate(x, 3, do_sample=False)
+    assert torch.equal(toks[:, 16], m(x)[:, -1].argmax(-1))
+    # loss path + backward
+    loss = m(x, labels=torch.randint(0, 256, (2, 16)))
+    loss.backward()
+    assert torch.isfinite(loss)
+
+
+def test_speculator_entry_bigcode(tmp_path):
+    from speculator import train_speculator as ts
+    ts.main(model_variant="gpt_bigcode_test", model_arch="embedgptbigcode",
+            use_dummy_dataset=True, batch_size=1, seq_length=128,
+            num_steps=1, report_interval=1, checkpoint_interval=100,
+            mixed_precision=False, n_speculator_heads=2, speculator_width=64,
+            stage2_start_step=10, model_path="/nonexistent",
+            ckpt_save_path=str(tmp_path), ckpt_load_path=str(tmp_path),
+            vocab_size=256, learning_rate=1e-4, sharding_strategy="fsdp")
