@@ -59,17 +59,31 @@ def main():
 
     torch.manual_seed(2023)
     mcfg = get_model_config(args.model)
-    with torch.device(device):
-        model = Llama(mcfg)
-        model.reset_parameters()
+    if args.model.startswith("mamba"):
+        from fms_fsdp_amd.models.mamba import (MambaBlock, MambaConfig,
+                                               MambaLMHeadModel)
+        mc = MambaConfig.from_dict(mcfg)
+        with torch.device(device):
+            model = MambaLMHeadModel(mc)
+            model.reset_parameters()
+        block_cls = MambaBlock
+
+        class _V:
+            src_vocab_size = mc.vocab_size
+        mcfg = _V()
+    else:
+        with torch.device(device):
+            model = Llama(mcfg)
+            model.reset_parameters()
+        block_cls = LlamaBlock
     n_params = model.param_count()
 
-    sm = ShardedModel(model, LlamaBlock, sharding_strategy=args.sharding,
+    sm = ShardedModel(model, block_cls, sharding_strategy=args.sharding,
                       param_dtype=torch.bfloat16 if use_cuda else torch.float32,
                       reshard_after_forward=n_params > 30e9,
                       prefetch_lookahead=1, device=device)
     if args.ac not in ("0", "0.0", 0):
-        apply_selective_ac(sm, LlamaBlock, args.ac)
+        apply_selective_ac(sm, block_cls, args.ac)
     opt = ShardedAdamW(sm, lr=3e-4)
 
     bs, sl = args.batch_size, args.seq_len
