@@ -158,10 +158,14 @@ def train(cfg, model, local_rank, rank, train_loader, optimizer, scheduler,
             if rank == 0:
                 mem = (torch.cuda.max_memory_allocated() / 2**30
                        if torch.cuda.is_available() else 0)
+                overall = elapsed_tokens / (time.time() - start) / world
+                tokens_per_day = elapsed_tokens / (time.time() - start) * 86400
                 print(f"step {batch_idx}: loss={train_loss:.4f} "
                       f"gnorm={train_gnorm:.3f} lr={scheduler.get_last_lr()[0]:.2e} "
                       f"tok/s/gpu={tok_per_gpu_sec:,.0f} "
+                      f"(overall {overall:,.0f}) "
                       f"tokens_seen={n_tok + elapsed_tokens:,} "
+                      f"tok/day={tokens_per_day:,.0f} "
                       f"peak_mem={mem:.1f}GiB")
                 train_result = {"loss": train_loss, "gnorm": train_gnorm,
                                 "tok_per_gpu_sec": tok_per_gpu_sec}
