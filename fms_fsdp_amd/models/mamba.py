@@ -180,14 +180,25 @@ class Mamba2Mixer(nn.Module):
             xBC, [self.d_inner,
                   self.ngroups * self.d_state, self.ngroups * self.d_state],
             dim=-1)
-        dtf = F.softplus(dt.float() + self.dt_bias.float())
-        A = -torch.exp(self.A_log.float())
-        y = ssd_chunked(
-            x.view(b, l, self.nheads, self.headdim).float(), dtf, A,
-            B.view(b, l, self.ngroups, self.d_state).float(),
-            C.view(b, l, self.ngroups, self.d_state).float(), self.chunk)
-        y = y + x.view(b, l, self.nheads, self.headdim).float() \
-            * self.D.view(1, 1, -1, 1)
+
+        def _scan(x_, dt_, dtb, Alog, B_, C_, D_):
+            # fp32 SSD; recomputed in backward (the chunked intermediates
+            # -- L, scores, states -- are large and cheap to rebuild)
+            dtf = F.softplus(dt_.float() + dtb.float())
+            A = -torch.exp(Alog.float())
+            xh = x_.view(b, l, self.nheads, self.headdim).float()
+            y_ = ssd_chunked(xh, dtf, A,
+                             B_.view(b, l, self.ngroups, self.d_state).float(),
+                             C_.view(b, l, self.ngroups, self.d_state).float(),
+                             self.chunk)
+            return y_ + xh * D_.view(1, 1, -1, 1)
+
+        if torch.is_grad_enabled() and self.training:
+            y = torch.utils.checkpoint.checkpoint(
+                _scan, x, dt, self.dt_bias, self.A_log, B, C, self.D,
+                use_reentrant=False)
+        else:
+            y = _scan(x, dt, self.dt_bias, self.A_log, B, C, self.D)
         y = y.reshape(b, l, self.d_inner).to(u.dtype)
         y = self.norm(y * F.silu(z))
         return self.out_proj(y)
