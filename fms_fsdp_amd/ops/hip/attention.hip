@@ -253,28 +253,57 @@ __global__ void attn_bwd_delta_kernel(const short* __restrict__ dog,
 }
 
 // ---------------------------------------------------------------------
-// Backward dq: grid like forward (q-tiles); recompute P and dP, then
+// Shared helpers for the backward kernels: SUBTILED LDS images.
+// A [R rows][D cols] bf16 tile is stored as [R/4][D/16][4][16] (each
+// subtile 128 B contiguous). This single layout serves BOTH fragment
+// patterns:
+//  - plain B/A reads (row = lane&31, 8 consecutive cols): 16 B contiguous
+//  - transposed reads via gfx950 ds_read_b64_tr_b16: a 16-lane group's
+//    lane j at subtile_base + j*8 receives COLUMN j of the 4x16 subtile
+//    (semantics verified on hardware: tools/tr_probe.hip).
+// ---------------------------------------------------------------------
+#define SUBT_OFF(row, col, DBLK) \
+  ((((row) >> 2) * (DBLK) + ((col) >> 4)) * 128 + ((row) & 3) * 32 + \
+   ((col) & 15) * 2)
+
+__device__ __forceinline__ unsigned long long tr_read(const char* base,
+                                                      int byte_off) {
+  unsigned long long v;
+  const unsigned addr = (unsigned)(unsigned long long)(base + byte_off);
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=v"(v)
+               : "v"(addr)
+               : "memory");
+  return v;
+}
+
+union U2x64 {
+  unsigned long long u[2];
+  bf16x8v v;
+};
+
+// ---------------------------------------------------------------------
+// Backward dq: grid over q-tiles; recompute P and dP once, then
 // dq^T = K^T @ dS^T. No atomics: each wave owns its 32 q rows.
-//   P^T  = exp(scale*K@Q^T - lse)          (A=K rows from k_lds)
-//   dP^T = V @ dO^T                        (A=V rows from v_lds)
+//   P^T  = exp(scale*K@Q^T - lse)    (A = K rows from the k image)
+//   dP^T = V @ dO^T                  (A = V rows from the v image)
 //   dS^T = P^T * (dP^T - delta) * scale
 //   dq^T[dk][q] = sum_key K^T[dk][key] dS^T[key][q]
-//                 (A=K^T from kt_lds, B=pack(dS^T))
+//     (A = K^T via tr_read on the SAME k image, B = pack(dS^T))
 // ---------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const short* __restrict__ dog, const short* __restrict__ qg,
     const short* __restrict__ kg, const short* __restrict__ vg,
     const float* __restrict__ lseg, const float* __restrict__ deltag,
     short* __restrict__ dqg, int B, int S, int H, int KVH, float scale) {
   constexpr int KVB = 32;
-  constexpr int KSWZ = (D == 128) ? 15 : 7;
   constexpr int NC = D / 16;
   constexpr int NT = D / 32;
+  constexpr int DBLK = D / 16;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_lds = smem;                       // [KVB][D*2] swizzled
-  char* v_lds = smem + KVB * D * 2;         // [KVB][D*2] swizzled
-  char* kt_lds = smem + 2 * KVB * D * 2;    // [D][KVB*2] transposed K
+  char* k_img = smem;                       // subtiled [KVB][D]
+  char* v_img = smem + KVB * D * 2;         // subtiled [KVB][D]
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -316,54 +345,31 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
   for (int tile = 0; tile < ntiles; ++tile) {
     const int kv0 = tile * KVB;
-    // stage K, V row-major (swizzled) + K transposed
+    // stage K/V into subtiled images: thread t: key = t>>3, d0 = (t&7)*16
     {
       const int t256 = threadIdx.x;
-      constexpr int BPR = D * 2;
-      constexpr int CHUNKS = KVB * BPR / 16;
-#pragma unroll
-      for (int i = t256; i < CHUNKS; i += 256) {
-        const int row = i / (BPR / 16);
-        const int cb = (i % (BPR / 16)) * 16;
-        const long long g = (long long)(kv0 + row) * krow_stride + cb / 2;
-        *(f32x4*)(k_lds + ((row * BPR + cb) ^ ((row & KSWZ) << 4))) =
-            *(const f32x4*)(kbase + g);
-        *(f32x4*)(v_lds + ((row * BPR + cb) ^ ((row & KSWZ) << 4))) =
-            *(const f32x4*)(vbase + g);
-      }
-      // K transposed: thread t: key = t % KVB, dk group = t/KVB
-      const int key = t256 % KVB;
-      const int ng = 256 / KVB;              // thread groups over dk
-      const int dk0 = (t256 / KVB) * (D / ng);
-      const short* kp = kbase + (long long)(kv0 + key) * krow_stride + dk0;
-#pragma unroll
-      for (int jj = 0; jj < D / ng / 8; ++jj) {
-        const bf16x8v kv8 = *(const bf16x8v*)(kp + jj * 8);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int dk = dk0 + jj * 8 + e;
-          *(__bf16*)(kt_lds + ((dk * (KVB * 2) + key * 2) ^ ((dk & 3) << 4))) =
-              kv8[e];
-        }
-      }
+      const int key = t256 >> 3;
+      const int d0 = (t256 & 7) * 16;
+      const long long g = (long long)(kv0 + key) * krow_stride + d0;
+      const int off = SUBT_OFF(key, d0, DBLK);
+      *(f32x4*)(k_img + off) = *(const f32x4*)(kbase + g);
+      *(f32x4*)(k_img + off + 16) = *(const f32x4*)(kbase + g + 8);
+      *(f32x4*)(v_img + off) = *(const f32x4*)(vbase + g);
+      *(f32x4*)(v_img + off + 16) = *(const f32x4*)(vbase + g + 8);
     }
     __syncthreads();
 
-    // S^T and dP^T (one 32-key tile)
+    // S^T and dP^T (A rows = keys, plain subtiled reads)
     f32x16 accS = (f32x16)(0.f), accDP = (f32x16)(0.f);
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
-      const int row = col;
-      const int inrow = c * 32 + hb * 16;
-      const bf16x8v ka =
-          *(const bf16x8v*)(k_lds + ((row * (D * 2) + inrow) ^ ((row & KSWZ) << 4)));
-      const bf16x8v va =
-          *(const bf16x8v*)(v_lds + ((row * (D * 2) + inrow) ^ ((row & KSWZ) << 4)));
+      const int off = SUBT_OFF(col, c * 16 + hb * 8, DBLK);
+      const bf16x8v ka = *(const bf16x8v*)(k_img + off);
+      const bf16x8v va = *(const bf16x8v*)(v_img + off);
       accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qb[c], accS, 0, 0, 0);
       accDP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dob[c], accDP, 0, 0, 0);
     }
 
-    // dS^T = P * (dP - delta) * scale  (0 where masked)
     float ds[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -374,26 +380,29 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       ds[r] = v;
     }
 
-    // dq^T += K^T @ dS^T
+    // dq^T += K^T @ dS^T : A = tr_read on k image (col dk across key rows)
     bf16x8v dsb[2];
     dsb[0] = pack_pT_chunk(ds);
     dsb[1] = pack_pT_chunk(ds + 8);
 #pragma unroll
     for (int t = 0; t < NT; ++t) {
+      const int dkblk = t * 2 + (col >> 4);
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
-        const int row = t * 32 + col;   // dk row
-        const int inrow = kc * 32 + hb * 16;
-        const bf16x8v a = *(const bf16x8v*)(
-            kt_lds + ((row * (KVB * 2) + inrow) ^ ((row & 3) << 4)));
+        const int key0 = kc * 16 + hb * 8;
+        U2x64 a;
+        a.u[0] = tr_read(k_img, ((key0 >> 2) * DBLK + dkblk) * 128 +
+                                    (lane & 15) * 8);
+        a.u[1] = tr_read(k_img, (((key0 + 4) >> 2) * DBLK + dkblk) * 128 +
+                                    (lane & 15) * 8);
         accDQ[t] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, dsb[kc], accDQ[t], 0, 0, 0);
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a.v, dsb[kc], accDQ[t],
+                                                    0, 0, 0);
       }
     }
     __syncthreads();
   }
 
-  // write dq (bf16): lane owns q row my_q; dq^T D-layout: dk = DROW + 32t
   short* dqp = dqg + ((long long)b * S * H + (long long)h) * D +
                (long long)my_q * qrow_stride;
 #pragma unroll
@@ -410,11 +419,13 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 }
 
 // ---------------------------------------------------------------------
-// Backward dk/dv: grid over KV tiles; each wave owns 32 keys, loops all
-// q-tiles >= its kv tile (causal). dK/dV accumulated in registers; P and
-// dS^T cross the lane<->reg transpose through a small LDS buffer.
-//   dV[key][dv] = sum_q P^T[key][q] dO[q][dv]   (A=P from p_lds, B=dOt)
-//   dK[key][dk] = sum_q dS^T[key][q] Q[q][dk]   (A=dS from p_lds, B=Qt)
+// Backward dk/dv: grid over KV tiles; each wave owns 32 keys and walks
+// all q-tiles >= its block's kv base. dK/dV accumulate in registers;
+// V stays in registers; K in a per-wave swizzled LDS tile; Q/dO arrive
+// per q-iteration in shared SUBTILED images (coalesced vector staging,
+// plain reads for the S/dP B-operands, tr_read for the dV/dK
+// B-operands). The P/dS lane<->reg transpose goes through a small
+// wave-private LDS buffer. TWO barriers per q-iteration.
 // ---------------------------------------------------------------------
 template <int D>
 __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
@@ -427,14 +438,12 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   constexpr int KSWZ = (D == 128) ? 15 : 7;
   constexpr int NC = D / 16;
   constexpr int NT = D / 32;
+  constexpr int DBLK = D / 16;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // per-wave carves
-  char* k_lds = smem;                                  // [4][KVB][D*2]
-  char* v_lds = smem + 4 * KVB * D * 2;                // [4][KVB][D*2]
-  // ONE shared transpose buffer, staged with dO^T then re-staged with Q^T
-  // each q-iteration: keeps total LDS at 80 KB -> 2 blocks/CU.
-  char* t_lds = smem + 8 * KVB * D * 2;                // [D][32*2] shared
-  char* p_lds = t_lds + D * 64;                        // [4][KVB][32*2]
+  char* k_lds = smem;                           // [4][KVB][D*2] per-wave
+  char* imgq = smem + 4 * KVB * D * 2;          // subtiled [32][D] shared
+  char* imgdo = imgq + 32 * D * 2;              // subtiled [32][D] shared
+  char* p_lds = imgdo + 32 * D * 2;             // [4][KVB][32*2]
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -446,7 +455,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   const int h = bh % H;
   const int kvh = h / (H / KVH);
   const int ngrp = H / KVH;
-  const int kv0 = blockIdx.x * 128 + wid * KVB;  // this wave's keys
+  const int kv0 = blockIdx.x * 128 + wid * KVB;
 
   const long long qrow_stride = (long long)H * D;
   const long long krow_stride = (long long)KVH * D;
@@ -456,23 +465,26 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   const short* vbase = vg + ((long long)b * S * KVH + (long long)kvh) * D;
 
   char* my_k = k_lds + wid * KVB * D * 2;
-  char* my_v = v_lds + wid * KVB * D * 2;
   char* my_p = p_lds + wid * KVB * 64;
 
-  // stage this wave's K/V rows once (each lane 32 16B chunks / wave)
+  // stage this wave's K rows once (swizzled row-major)
   {
     constexpr int BPR = D * 2;
     constexpr int CHUNKS = KVB * BPR / 16;
-#pragma unroll
     for (int i = lane; i < CHUNKS; i += 64) {
       const int row = i / (BPR / 16);
       const int cb = (i % (BPR / 16)) * 16;
       const long long g = (long long)(kv0 + row) * krow_stride + cb / 2;
       *(f32x4*)(my_k + ((row * BPR + cb) ^ ((row & KSWZ) << 4))) =
           *(const f32x4*)(kbase + g);
-      *(f32x4*)(my_v + ((row * BPR + cb) ^ ((row & KSWZ) << 4))) =
-          *(const f32x4*)(vbase + g);
     }
+  }
+  // V fragments stay in registers (A-operand rows = this lane's key)
+  bf16x8v vreg[NC];
+  {
+    const short* vp = vbase + (long long)(kv0 + col) * krow_stride + hb * 8;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) vreg[c] = *(const bf16x8v*)(vp + c * 16);
   }
 
   f32x16 accDV[NT], accDK[NT];
@@ -482,56 +494,37 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     accDK[t] = (f32x16)(0.f);
   }
 
-  // q tiles: causal => q >= kv0 of the BLOCK's first wave; all waves walk
-  // the same q range (block-uniform barriers), masking handles the rest.
   const int q_start = (blockIdx.x * 128) / 32 * 32;
   for (int q0 = q_start; q0 < S; q0 += 32) {
-    // stage dO^T (transposed, shared): thread t: q = t&31, dk grp t>>5
+    // stage Q and dO subtiled images (coalesced; no transpose writes)
     {
       const int t256 = threadIdx.x;
-      const int q = t256 & 31;
-      const int ng = 256 / 32;                  // 8 groups over dk
-      const int dk0 = (t256 >> 5) * (D / ng);
-      const short* dp = dobase + (long long)(q0 + q) * qrow_stride + dk0;
-#pragma unroll
-      for (int jj = 0; jj < D / ng / 8; ++jj) {
-        const bf16x8v dv8 = *(const bf16x8v*)(dp + jj * 8);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int dk = dk0 + jj * 8 + e;
-          *(__bf16*)(t_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = dv8[e];
-        }
-      }
+      const int q = t256 >> 3;
+      const int d0 = (t256 & 7) * 16;
+      const long long g = (long long)(q0 + q) * qrow_stride + d0;
+      const int off = SUBT_OFF(q, d0, DBLK);
+      *(f32x4*)(imgq + off) = *(const f32x4*)(qbase + g);
+      *(f32x4*)(imgq + off + 16) = *(const f32x4*)(qbase + g + 8);
+      *(f32x4*)(imgdo + off) = *(const f32x4*)(dobase + g);
+      *(f32x4*)(imgdo + off + 16) = *(const f32x4*)(dobase + g + 8);
     }
     __syncthreads();
 
-    // S^T, dP^T for (my 32 keys) x (32 q)
+    // S^T (A=K lds, B=imgq) and dP^T (A=V regs, B=imgdo)
     f32x16 accS = (f32x16)(0.f), accDP = (f32x16)(0.f);
-    // B operands: Q / dO rows q0+col, read from qt/dot (B[k=dk][j=q]):
-    // lane: dk=(l>>5)*8+e, q=l&31 -> qt[dk][q] strided... use global regs:
-    bf16x8v qb2[NC], dob2[NC];
-    {
-      const short* qp = qbase + (long long)(q0 + col) * qrow_stride + hb * 8;
-      const short* dp = dobase + (long long)(q0 + col) * qrow_stride + hb * 8;
-#pragma unroll
-      for (int c = 0; c < NC; ++c) {
-        qb2[c] = *(const bf16x8v*)(qp + c * 16);
-        dob2[c] = *(const bf16x8v*)(dp + c * 16);
-      }
-    }
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
-      const int row = col;
       const int inrow = c * 32 + hb * 16;
       const bf16x8v ka =
-          *(const bf16x8v*)(my_k + ((row * (D * 2) + inrow) ^ ((row & KSWZ) << 4)));
-      const bf16x8v va =
-          *(const bf16x8v*)(my_v + ((row * (D * 2) + inrow) ^ ((row & KSWZ) << 4)));
-      accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qb2[c], accS, 0, 0, 0);
-      accDP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dob2[c], accDP, 0, 0, 0);
+          *(const bf16x8v*)(my_k + ((col * (D * 2) + inrow) ^ ((col & KSWZ) << 4)));
+      const int boff = SUBT_OFF(col, c * 16 + hb * 8, DBLK);
+      const bf16x8v qbf = *(const bf16x8v*)(imgq + boff);
+      const bf16x8v dbf = *(const bf16x8v*)(imgdo + boff);
+      accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qbf, accS, 0, 0, 0);
+      accDP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vreg[c], dbf, accDP,
+                                                      0, 0, 0);
     }
 
-    // P^T and dS^T; write P to p_lds for the A-operand transpose
     const float lse_q = lseg[((long long)bh) * S + q0 + col];
     const float delta_q = deltag[((long long)bh) * S + q0 + col];
     float pv[16], ds[16];
@@ -544,80 +537,61 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       pv[r] = pval;
       ds[r] = pval * (accDP[r] - delta_q) * scale;
     }
-    // p_lds layout [key][q], rows 64B, swizzle ((key&3)<<4) — write P
+    // P -> my_p (wave-private; in-wave LDS ordering suffices)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key = DROW(r, hb);
       *(short*)(my_p + ((key * 64 + col * 2) ^ ((key & 3) << 4))) = f2bf(pv[r]);
     }
-    // (p_lds is wave-private: lgkmcnt ordering suffices, no barrier)
-    // dV[key][dv] += P(A) @ dOt(B): A[i=key][k=q] from p_lds
+    // dV[key][dv] += P(A) @ tr(imgdo)(B)
     {
       bf16x8v pa[2];
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
-        const int row = col;  // key row
         const int inrow = kc * 32 + hb * 16;
-        pa[kc] = *(const bf16x8v*)(my_p + ((row * 64 + inrow) ^ ((row & 3) << 4)));
+        pa[kc] = *(const bf16x8v*)(my_p + ((col * 64 + inrow) ^ ((col & 3) << 4)));
       }
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
+        const int dvblk = t * 2 + (col >> 4);
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
-          // B[k=q][j=dv] from dot_lds[dv... B read: lane: q=(l>>5)*8+e,
-          // dv=l&31: dot_lds holds [dk][q] (transposed dO): B[k=q][j=dv]
-          // = dO[q][dv] = dot_lds[dv][q] -> lane reads row dv=col... but
-          // j=l&31 must be dv: row = t*32+col, q chunk = kc*32+hb*16
-          const int row = t * 32 + col;
-          const int inrow = kc * 32 + hb * 16;
-          const bf16x8v bb = *(const bf16x8v*)(
-              t_lds + ((row * 64 + inrow) ^ ((row & 3) << 4)));
-          accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kc], bb,
+          const int qg0 = kc * 16 + hb * 8;
+          U2x64 bb;
+          bb.u[0] = tr_read(imgdo, ((qg0 >> 2) * DBLK + dvblk) * 128 +
+                                       (lane & 15) * 8);
+          bb.u[1] = tr_read(imgdo, (((qg0 + 4) >> 2) * DBLK + dvblk) * 128 +
+                                       (lane & 15) * 8);
+          accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kc], bb.v,
                                                              accDV[t], 0, 0, 0);
         }
       }
     }
-    __syncthreads();  // every wave done with the dO image
-    // re-stage t_lds with Q^T; overwrite p_lds with dS^T
-    {
-      const int t256 = threadIdx.x;
-      const int q = t256 & 31;
-      const int ng = 256 / 32;
-      const int dk0 = (t256 >> 5) * (D / ng);
-      const short* qp = qbase + (long long)(q0 + q) * qrow_stride + dk0;
-#pragma unroll
-      for (int jj = 0; jj < D / ng / 8; ++jj) {
-        const bf16x8v qv8 = *(const bf16x8v*)(qp + jj * 8);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int dk = dk0 + jj * 8 + e;
-          *(__bf16*)(t_lds + ((dk * 64 + q * 2) ^ ((dk & 3) << 4))) = qv8[e];
-        }
-      }
-    }
+    // dS -> my_p, then dK[key][dk] += dS(A) @ tr(imgq)(B)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key = DROW(r, hb);
       *(short*)(my_p + ((key * 64 + col * 2) ^ ((key & 3) << 4))) = f2bf(ds[r]);
     }
-    __syncthreads();
     {
       bf16x8v da[2];
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
-        const int row = col;
         const int inrow = kc * 32 + hb * 16;
-        da[kc] = *(const bf16x8v*)(my_p + ((row * 64 + inrow) ^ ((row & 3) << 4)));
+        da[kc] = *(const bf16x8v*)(my_p + ((col * 64 + inrow) ^ ((col & 3) << 4)));
       }
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
+        const int dkblk = t * 2 + (col >> 4);
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
-          const int row = t * 32 + col;
-          const int inrow = kc * 32 + hb * 16;
-          const bf16x8v bb = *(const bf16x8v*)(
-              t_lds + ((row * 64 + inrow) ^ ((row & 3) << 4)));
-          accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[kc], bb,
+          const int qg0 = kc * 16 + hb * 8;
+          U2x64 bb;
+          bb.u[0] = tr_read(imgq, ((qg0 >> 2) * DBLK + dkblk) * 128 +
+                                      (lane & 15) * 8);
+          bb.u[1] = tr_read(imgq, (((qg0 + 4) >> 2) * DBLK + dkblk) * 128 +
+                                      (lane & 15) * 8);
+          accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[kc], bb.v,
                                                              accDK[t], 0, 0, 0);
         }
       }
@@ -649,7 +623,6 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   }
 }
 
-// ---------------------------------------------------------------------
 extern "C" {
 
 void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
@@ -676,20 +649,20 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
       (const short*)do_, (const short*)o, delta_ws, D, S, H, rows);
   dim3 grid(S / 128, B * H);
   if (D == 128) {
-    const int lds_dq = 2 * 32 * 128 * 2 + 128 * 64;   // k + v + kt
+    const int lds_dq = 2 * 32 * 128 * 2;   // subtiled k + v images
     attn_bwd_dq_kernel<128><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
-    const int lds_dkv = 8 * 32 * 128 * 2 + 128 * 64 + 4 * 32 * 64;
+    const int lds_dkv = 4 * 32 * 128 * 2 + 2 * 32 * 128 * 2 + 4 * 32 * 64;
     attn_bwd_dkv_kernel<128><<<grid, 256, lds_dkv, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (float*)dk, (float*)dv, B, S, H, KVH, scale);
   } else {
-    const int lds_dq = 2 * 32 * 64 * 2 + 64 * 64;
+    const int lds_dq = 2 * 32 * 64 * 2;
     attn_bwd_dq_kernel<64><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
-    const int lds_dkv = 8 * 32 * 64 * 2 + 64 * 64 + 4 * 32 * 64;
+    const int lds_dkv = 4 * 32 * 64 * 2 + 2 * 32 * 64 * 2 + 4 * 32 * 64;
     attn_bwd_dkv_kernel<64><<<grid, 256, lds_dkv, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (float*)dk, (float*)dv, B, S, H, KVH, scale);
