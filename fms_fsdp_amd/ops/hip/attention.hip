@@ -266,21 +266,27 @@ __global__ void attn_bwd_delta_kernel(const short* __restrict__ dog,
   ((((row) >> 2) * (DBLK) + ((col) >> 4)) * 128 + ((row) & 3) * 32 + \
    ((col) & 15) * 2)
 
-__device__ __forceinline__ unsigned long long tr_read(const char* base,
-                                                      int byte_off) {
-  unsigned long long v;
-  const unsigned addr = (unsigned)(unsigned long long)(base + byte_off);
-  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
-               : "=v"(v)
-               : "v"(addr)
-               : "memory");
-  return v;
-}
-
 union U2x64 {
   unsigned long long u[2];
   bf16x8v v;
 };
+
+// Two transpose reads batched behind ONE lgkmcnt wait (each read's 4
+// shorts = column lane&15 of the 4x16 subtile at its address).
+__device__ __forceinline__ bf16x8v tr_read2(const char* base, int off0,
+                                            int off1) {
+  U2x64 r;
+  const unsigned a0 = (unsigned)(unsigned long long)(base + off0);
+  const unsigned a1 = (unsigned)(unsigned long long)(base + off1);
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(r.u[0]), "=&v"(r.u[1])
+      : "v"(a0), "v"(a1)
+      : "memory");
+  return r.v;
+}
 
 // ---------------------------------------------------------------------
 // Backward dq: grid over q-tiles; recompute P and dP once, then
@@ -390,13 +396,12 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         const int key0 = kc * 16 + hb * 8;
-        U2x64 a;
-        a.u[0] = tr_read(k_img, ((key0 >> 2) * DBLK + dkblk) * 128 +
-                                    (lane & 15) * 8);
-        a.u[1] = tr_read(k_img, (((key0 + 4) >> 2) * DBLK + dkblk) * 128 +
-                                    (lane & 15) * 8);
+        const bf16x8v a = tr_read2(
+            k_img,
+            ((key0 >> 2) * DBLK + dkblk) * 128 + (lane & 15) * 8,
+            (((key0 + 4) >> 2) * DBLK + dkblk) * 128 + (lane & 15) * 8);
         accDQ[t] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a.v, dsb[kc], accDQ[t],
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, dsb[kc], accDQ[t],
                                                     0, 0, 0);
       }
     }
@@ -557,12 +562,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
           const int qg0 = kc * 16 + hb * 8;
-          U2x64 bb;
-          bb.u[0] = tr_read(imgdo, ((qg0 >> 2) * DBLK + dvblk) * 128 +
-                                       (lane & 15) * 8);
-          bb.u[1] = tr_read(imgdo, (((qg0 + 4) >> 2) * DBLK + dvblk) * 128 +
-                                       (lane & 15) * 8);
-          accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kc], bb.v,
+          const bf16x8v bb = tr_read2(
+              imgdo,
+              ((qg0 >> 2) * DBLK + dvblk) * 128 + (lane & 15) * 8,
+              (((qg0 + 4) >> 2) * DBLK + dvblk) * 128 + (lane & 15) * 8);
+          accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kc], bb,
                                                              accDV[t], 0, 0, 0);
         }
       }
@@ -586,12 +590,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
           const int qg0 = kc * 16 + hb * 8;
-          U2x64 bb;
-          bb.u[0] = tr_read(imgq, ((qg0 >> 2) * DBLK + dkblk) * 128 +
-                                      (lane & 15) * 8);
-          bb.u[1] = tr_read(imgq, (((qg0 + 4) >> 2) * DBLK + dkblk) * 128 +
-                                      (lane & 15) * 8);
-          accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[kc], bb.v,
+          const bf16x8v bb = tr_read2(
+              imgq,
+              ((qg0 >> 2) * DBLK + dkblk) * 128 + (lane & 15) * 8,
+              (((qg0 + 4) >> 2) * DBLK + dkblk) * 128 + (lane & 15) * 8);
+          accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[kc], bb,
                                                              accDK[t], 0, 0, 0);
         }
       }
