@@ -65,8 +65,13 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   constexpr int NC = D / 16;   // QK^T k-chunks
   constexpr int NT = D / 32;   // 32-wide output tiles
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_lds = smem;                    // [KVB][D*2] swizzled, 256B rows @D=128
-  char* vt_lds = smem + KVB * D * 2;     // [D][KVB*2] transposed V, 128B rows
+  // double-buffered: stage tile t+1 while computing tile t, ONE barrier
+  // per tile (the end-of-iter barrier both publishes buf^1 and confirms
+  // every wave is done reading buf before iter t+2 overwrites it)
+  constexpr int KB = KVB * D * 2;
+  constexpr int VB = D * KVB * 2;
+#define KLDS(buf) (smem + ((buf) ? (KB + VB) : 0))
+#define VTLDS(buf) (smem + KB + ((buf) ? (KB + VB) : 0))
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -102,40 +107,43 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   float m_run = -1e30f, l_run = 0.f;
 
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
-  for (int tile = 0; tile < ntiles; ++tile) {
-    const int kv0 = tile * KVB;
-    // ---- cooperative stage: K row-major (swizzled), V transposed ----
-    {
-      // K: 256 threads, each copies KVB*D*2/256 bytes in 16B units
-      const int t256 = threadIdx.x;
-      constexpr int BYTES_PER_ROW = D * 2;
-      constexpr int CHUNKS = KVB * BYTES_PER_ROW / 16;  // 16B chunks
+  const int t256 = threadIdx.x;
+  auto stage = [&](int buf, int kv0) {
+    // K: 256 threads, each copies KVB*D*2/256 bytes in 16B units
+    constexpr int BYTES_PER_ROW = D * 2;
+    constexpr int CHUNKS = KVB * BYTES_PER_ROW / 16;  // 16B chunks
 #pragma unroll
-      for (int i = t256; i < CHUNKS; i += 256) {
-        const int row = i / (BYTES_PER_ROW / 16);
-        const int cb = (i % (BYTES_PER_ROW / 16)) * 16;
-        const long long g = (long long)(kv0 + row) * krow_stride + cb / 2;
-        *(f32x4*)(k_lds + ((row * BYTES_PER_ROW + cb) ^ ((row & KSWZ) << 4))) =
-            *(const f32x4*)(kbase + g);
-      }
-      // V transposed: thread t: key = t&63, dv block = (t>>6)*32.
-      // Loads are vectorized 16B (guide G13: never scalar bf16 loads);
-      // the transposed LDS writes scatter but stay cheap vs the MFMAs.
-      const int key = t256 & 63;
-      const int dv0 = (t256 >> 6) * (D / 4);
-      const short* vp = vbase + (long long)(kv0 + key) * krow_stride + dv0;
+    for (int i = t256; i < CHUNKS; i += 256) {
+      const int row = i / (BYTES_PER_ROW / 16);
+      const int cb = (i % (BYTES_PER_ROW / 16)) * 16;
+      const long long g = (long long)(kv0 + row) * krow_stride + cb / 2;
+      *(f32x4*)(KLDS(buf) +
+                ((row * BYTES_PER_ROW + cb) ^ ((row & KSWZ) << 4))) =
+          *(const f32x4*)(kbase + g);
+    }
+    // V transposed: thread t: key = t&63, dv block = (t>>6)*32.
+    // Loads vectorized 16B (guide G13); transposed writes scatter but
+    // stay cheap vs the MFMAs.
+    const int key = t256 & 63;
+    const int dv0 = (t256 >> 6) * (D / 4);
+    const short* vp = vbase + (long long)(kv0 + key) * krow_stride + dv0;
 #pragma unroll
-      for (int jj = 0; jj < D / 32; ++jj) {
-        const bf16x8v vv = *(const bf16x8v*)(vp + jj * 8);
+    for (int jj = 0; jj < D / 32; ++jj) {
+      const bf16x8v vv = *(const bf16x8v*)(vp + jj * 8);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int dv = dv0 + jj * 8 + e;
-          *(__bf16*)(vt_lds +
-                     ((dv * (KVB * 2) + key * 2) ^ ((dv & 7) << 4))) = vv[e];
-        }
+      for (int e = 0; e < 8; ++e) {
+        const int dv = dv0 + jj * 8 + e;
+        *(__bf16*)(VTLDS(buf) +
+                   ((dv * (KVB * 2) + key * 2) ^ ((dv & 7) << 4))) = vv[e];
       }
     }
-    __syncthreads();
+  };
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int tile = 0; tile < ntiles; ++tile) {
+    const int kv0 = tile * KVB;
+    if (tile + 1 < ntiles) stage(cur ^ 1, (tile + 1) * KVB);
 
     // ---- two 32-key sub-tiles, each with its own online-softmax pass.
     // Register economy: one live accS/p set (16 regs) instead of two,
@@ -153,8 +161,8 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       for (int c = 0; c < NC; ++c) {
         const int row = kt * 32 + col;
         const int inrow = c * 32 + hb * 16;
-        const bf16x8v a =
-            *(const bf16x8v*)(k_lds + ((row * (D * 2) + inrow) ^ ((row & KSWZ) << 4)));
+        const bf16x8v a = *(const bf16x8v*)(
+            KLDS(cur) + ((row * (D * 2) + inrow) ^ ((row & KSWZ) << 4)));
         accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qb[c], accS, 0, 0, 0);
       }
       float p[16];
@@ -193,17 +201,18 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
         const int row = t * 32 + col;   // dv row
         const int inrow0 = (kt * 2) * 32 + hb * 16;
         const bf16x8v a0 = *(const bf16x8v*)(
-            vt_lds + ((row * (KVB * 2) + inrow0) ^ ((row & 7) << 4)));
+            VTLDS(cur) + ((row * (KVB * 2) + inrow0) ^ ((row & 7) << 4)));
         accO[t] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, pb0, accO[t], 0, 0, 0);
         const int inrow1 = (kt * 2 + 1) * 32 + hb * 16;
         const bf16x8v a1 = *(const bf16x8v*)(
-            vt_lds + ((row * (KVB * 2) + inrow1) ^ ((row & 7) << 4)));
+            VTLDS(cur) + ((row * (KVB * 2) + inrow1) ^ ((row & 7) << 4)));
         accO[t] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, pb1, accO[t], 0, 0, 0);
       }
     }
     __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: o[q][dv] = accO^T / l ----
@@ -225,6 +234,8 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   if (hb == 0)
     lseg[((long long)bh) * S + my_q] = m_run + __logf(l_run);
 }
+#undef KLDS
+#undef VTLDS
 
 // ---------------------------------------------------------------------
 // Backward preprocess: delta[b,h,s] = rowsum(dO * O) fp32
@@ -632,7 +643,7 @@ void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
                      float* lse, int B, int S, int H, int KVH, int D,
                      float scale, hipStream_t stream) {
   dim3 grid(S / 128, B * H);
-  const int lds = 64 * D * 2 + D * 64 * 2;  // K + Vt
+  const int lds = 2 * (64 * D * 2 + D * 64 * 2);  // dbuf K + Vt
   if (D == 128)
     attn_fwd_kernel<128><<<grid, 256, lds, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse, B,
