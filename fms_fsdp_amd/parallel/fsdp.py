@@ -259,6 +259,10 @@ class ShardedModel(nn.Module):
                 shard_group, replicate_group = mesh_shard, mesh_rep
                 if dist.get_world_size(shard_group) == 1:
                     shard_group = None
+                # single node: pure intra-node sharding, no replica dim
+                if replicate_group is not None and \
+                        dist.get_world_size(replicate_group) == 1:
+                    replicate_group = None
             elif sharding_strategy == "ddp":
                 shard_group = None
                 replicate_group = dist.group.WORLD
