@@ -273,9 +273,13 @@ __global__ void attn_bwd_delta_kernel(const short* __restrict__ dog,
 //    lane j at subtile_base + j*8 receives COLUMN j of the 4x16 subtile
 //    (semantics verified on hardware: tools/tr_probe.hip).
 // ---------------------------------------------------------------------
+// column-block-major subtiles: tile = colblk*(R/4=8) + rowgrp. A plain
+// 16-lane-group read then touches 4 CONSECUTIVE tiles (alternating LDS
+// bank halves); the rowgrp-bit1 16B XOR spreads the remaining overlap.
+// (PMC-driven: the row-group-major variant was a 4-way conflict.)
 #define SUBT_OFF(row, col, DBLK) \
-  ((((row) >> 2) * (DBLK) + ((col) >> 4)) * 128 + ((row) & 3) * 32 + \
-   ((col) & 15) * 2)
+  ((((col) >> 4) * 8 + ((row) >> 2)) * 128 + \
+   ((((row) & 3) * 32 + ((col) & 15) * 2) ^ ((((row) >> 2) & 2) << 3)))
 
 union U2x64 {
   unsigned long long u[2];
@@ -364,15 +368,18 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const int kv0 = tile * KVB;
     // stage K/V into subtiled images: thread t: key = t>>3, d0 = (t&7)*16
     {
+      // thread t: key = t&31, colblk = t>>5 (8-lane write groups span
+      // adjacent rowgrp tiles -> conflict-free banks)
       const int t256 = threadIdx.x;
-      const int key = t256 >> 3;
-      const int d0 = (t256 & 7) * 16;
+      const int key = t256 & 31;
+      const int d0 = (t256 >> 5) * 16;
       const long long g = (long long)(kv0 + key) * krow_stride + d0;
-      const int off = SUBT_OFF(key, d0, DBLK);
-      *(f32x4*)(k_img + off) = *(const f32x4*)(kbase + g);
-      *(f32x4*)(k_img + off + 16) = *(const f32x4*)(kbase + g + 8);
-      *(f32x4*)(v_img + off) = *(const f32x4*)(vbase + g);
-      *(f32x4*)(v_img + off + 16) = *(const f32x4*)(vbase + g + 8);
+      const int off0 = SUBT_OFF(key, d0, DBLK);
+      const int off1 = SUBT_OFF(key, d0 + 8, DBLK);
+      *(f32x4*)(k_img + off0) = *(const f32x4*)(kbase + g);
+      *(f32x4*)(k_img + off1) = *(const f32x4*)(kbase + g + 8);
+      *(f32x4*)(v_img + off0) = *(const f32x4*)(vbase + g);
+      *(f32x4*)(v_img + off1) = *(const f32x4*)(vbase + g + 8);
     }
     __syncthreads();
 
@@ -407,10 +414,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         const int key0 = kc * 16 + hb * 8;
+        const int rg0 = key0 >> 2, rg1 = (key0 + 4) >> 2;
         const bf16x8v a = tr_read2(
             k_img,
-            ((key0 >> 2) * DBLK + dkblk) * 128 + (lane & 15) * 8,
-            (((key0 + 4) >> 2) * DBLK + dkblk) * 128 + (lane & 15) * 8);
+            (dkblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
+            (dkblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
         accDQ[t] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, dsb[kc], accDQ[t],
                                                     0, 0, 0);
@@ -515,14 +523,15 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     // stage Q and dO subtiled images (coalesced; no transpose writes)
     {
       const int t256 = threadIdx.x;
-      const int q = t256 >> 3;
-      const int d0 = (t256 & 7) * 16;
+      const int q = t256 & 31;
+      const int d0 = (t256 >> 5) * 16;
       const long long g = (long long)(q0 + q) * qrow_stride + d0;
-      const int off = SUBT_OFF(q, d0, DBLK);
-      *(f32x4*)(imgq + off) = *(const f32x4*)(qbase + g);
-      *(f32x4*)(imgq + off + 16) = *(const f32x4*)(qbase + g + 8);
-      *(f32x4*)(imgdo + off) = *(const f32x4*)(dobase + g);
-      *(f32x4*)(imgdo + off + 16) = *(const f32x4*)(dobase + g + 8);
+      const int off0 = SUBT_OFF(q, d0, DBLK);
+      const int off1 = SUBT_OFF(q, d0 + 8, DBLK);
+      *(f32x4*)(imgq + off0) = *(const f32x4*)(qbase + g);
+      *(f32x4*)(imgq + off1) = *(const f32x4*)(qbase + g + 8);
+      *(f32x4*)(imgdo + off0) = *(const f32x4*)(dobase + g);
+      *(f32x4*)(imgdo + off1) = *(const f32x4*)(dobase + g + 8);
     }
     __syncthreads();
 
@@ -573,10 +582,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
           const int qg0 = kc * 16 + hb * 8;
+          const int rg0 = qg0 >> 2, rg1 = (qg0 + 4) >> 2;
           const bf16x8v bb = tr_read2(
               imgdo,
-              ((qg0 >> 2) * DBLK + dvblk) * 128 + (lane & 15) * 8,
-              (((qg0 + 4) >> 2) * DBLK + dvblk) * 128 + (lane & 15) * 8);
+              (dvblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
+              (dvblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
           accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kc], bb,
                                                              accDV[t], 0, 0, 0);
         }
@@ -601,10 +611,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
           const int qg0 = kc * 16 + hb * 8;
+          const int rg0 = qg0 >> 2, rg1 = (qg0 + 4) >> 2;
           const bf16x8v bb = tr_read2(
               imgq,
-              ((qg0 >> 2) * DBLK + dkblk) * 128 + (lane & 15) * 8,
-              (((qg0 + 4) >> 2) * DBLK + dkblk) * 128 + (lane & 15) * 8);
+              (dkblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
+              (dkblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
           accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[kc], bb,
                                                              accDK[t], 0, 0, 0);
         }
