@@ -297,10 +297,13 @@ class MambaBlock(nn.Module):
         return self._forward_impl(x)
 
     def _forward_impl(self, x):
-        x = x + self.mixer(self.norm(x))
+        h = self.mixer(self.norm(x))
         if self.mlp is not None:
-            x = x + self.mlp(self.norm2(x))
-        return x
+            # fused residual-add + norm (the reference's fused_add_norm,
+            # config_utils.py:182 there)
+            y2, s = ops.add_rmsnorm(x, h, self.norm2.weight, self.norm2.eps)
+            return s + self.mlp(y2)
+        return x + h
 
 
 class MambaLMHeadModel(nn.Module):
