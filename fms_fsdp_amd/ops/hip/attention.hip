@@ -323,8 +323,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   constexpr int NT = D / 32;
   constexpr int DBLK = D / 16;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_img = smem;                       // subtiled [KVB][D]
-  char* v_img = smem + KVB * D * 2;         // subtiled [KVB][D]
+  // double-buffered subtiled K/V images: stage tile t+1 during tile t's
+  // compute, ONE barrier per tile.
+  constexpr int IMGB = KVB * D * 2;
+#define KIMG(buf) (smem + ((buf) ? 2 * IMGB : 0))
+#define VIMG(buf) (smem + IMGB + ((buf) ? 2 * IMGB : 0))
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -364,32 +367,34 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   for (int t = 0; t < NT; ++t) accDQ[t] = (f32x16)(0.f);
 
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
+  const int t256 = threadIdx.x;
+  // stage K/V into subtiled images: thread t: key = t&31, colblk = t>>5
+  // (8-lane write groups span adjacent rowgrp tiles -> conflict-free)
+  auto stage = [&](int buf, int kv0s) {
+    const int key = t256 & 31;
+    const int d0 = (t256 >> 5) * 16;
+    const long long g = (long long)(kv0s + key) * krow_stride + d0;
+    const int off0 = SUBT_OFF(key, d0, DBLK);
+    const int off1 = SUBT_OFF(key, d0 + 8, DBLK);
+    *(f32x4*)(KIMG(buf) + off0) = *(const f32x4*)(kbase + g);
+    *(f32x4*)(KIMG(buf) + off1) = *(const f32x4*)(kbase + g + 8);
+    *(f32x4*)(VIMG(buf) + off0) = *(const f32x4*)(vbase + g);
+    *(f32x4*)(VIMG(buf) + off1) = *(const f32x4*)(vbase + g + 8);
+  };
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
   for (int tile = 0; tile < ntiles; ++tile) {
+    if (tile + 1 < ntiles) stage(cur ^ 1, (tile + 1) * KVB);
     const int kv0 = tile * KVB;
-    // stage K/V into subtiled images: thread t: key = t>>3, d0 = (t&7)*16
-    {
-      // thread t: key = t&31, colblk = t>>5 (8-lane write groups span
-      // adjacent rowgrp tiles -> conflict-free banks)
-      const int t256 = threadIdx.x;
-      const int key = t256 & 31;
-      const int d0 = (t256 >> 5) * 16;
-      const long long g = (long long)(kv0 + key) * krow_stride + d0;
-      const int off0 = SUBT_OFF(key, d0, DBLK);
-      const int off1 = SUBT_OFF(key, d0 + 8, DBLK);
-      *(f32x4*)(k_img + off0) = *(const f32x4*)(kbase + g);
-      *(f32x4*)(k_img + off1) = *(const f32x4*)(kbase + g + 8);
-      *(f32x4*)(v_img + off0) = *(const f32x4*)(vbase + g);
-      *(f32x4*)(v_img + off1) = *(const f32x4*)(vbase + g + 8);
-    }
-    __syncthreads();
 
     // S^T and dP^T (A rows = keys, plain subtiled reads)
     f32x16 accS = (f32x16)(0.f), accDP = (f32x16)(0.f);
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
       const int off = SUBT_OFF(col, c * 16 + hb * 8, DBLK);
-      const bf16x8v ka = *(const bf16x8v*)(k_img + off);
-      const bf16x8v va = *(const bf16x8v*)(v_img + off);
+      const bf16x8v ka = *(const bf16x8v*)(KIMG(cur) + off);
+      const bf16x8v va = *(const bf16x8v*)(VIMG(cur) + off);
       accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qb[c], accS, 0, 0, 0);
       accDP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dob[c], accDP, 0, 0, 0);
     }
@@ -416,7 +421,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
         const int key0 = kc * 16 + hb * 8;
         const int rg0 = key0 >> 2, rg1 = (key0 + 4) >> 2;
         const bf16x8v a = tr_read2(
-            k_img,
+            KIMG(cur),
             (dkblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
             (dkblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
         accDQ[t] =
@@ -425,6 +430,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       }
     }
     __syncthreads();
+    cur ^= 1;
   }
 
   short* dqp = dqg + ((long long)b * S * H + (long long)h) * D +
@@ -441,6 +447,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     }
   }
 }
+#undef KIMG
+#undef VIMG
 
 // ---------------------------------------------------------------------
 // Backward dk/dv: grid over KV tiles; each wave owns 32 keys and walks
@@ -465,9 +473,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   constexpr int DBLK = D / 16;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                           // [4][KVB][D*2] per-wave
-  char* imgq = smem + 4 * KVB * D * 2;          // subtiled [32][D] shared
-  char* imgdo = imgq + 32 * D * 2;              // subtiled [32][D] shared
-  char* p_lds = imgdo + 32 * D * 2;             // [4][KVB][32*2]
+  // double-buffered subtiled Q/dO images (stage q-tile t+1 during t)
+  constexpr int IMGB = 32 * D * 2;
+#define QIMG(buf) (smem + 4 * KVB * D * 2 + ((buf) ? 2 * IMGB : 0))
+#define DOIMG(buf) (smem + 4 * KVB * D * 2 + IMGB + ((buf) ? 2 * IMGB : 0))
+  char* p_lds = smem + 4 * KVB * D * 2 + 4 * IMGB;  // [4][KVB][32*2]
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -519,21 +529,23 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   }
 
   const int q_start = (blockIdx.x * 128) / 32 * 32;
+  const int t256 = threadIdx.x;
+  auto stage = [&](int buf, int q0s) {
+    const int q = t256 & 31;
+    const int d0 = (t256 >> 5) * 16;
+    const long long g = (long long)(q0s + q) * qrow_stride + d0;
+    const int off0 = SUBT_OFF(q, d0, DBLK);
+    const int off1 = SUBT_OFF(q, d0 + 8, DBLK);
+    *(f32x4*)(QIMG(buf) + off0) = *(const f32x4*)(qbase + g);
+    *(f32x4*)(QIMG(buf) + off1) = *(const f32x4*)(qbase + g + 8);
+    *(f32x4*)(DOIMG(buf) + off0) = *(const f32x4*)(dobase + g);
+    *(f32x4*)(DOIMG(buf) + off1) = *(const f32x4*)(dobase + g + 8);
+  };
+  stage(0, q_start);
+  __syncthreads();
+  int cur = 0;
   for (int q0 = q_start; q0 < S; q0 += 32) {
-    // stage Q and dO subtiled images (coalesced; no transpose writes)
-    {
-      const int t256 = threadIdx.x;
-      const int q = t256 & 31;
-      const int d0 = (t256 >> 5) * 16;
-      const long long g = (long long)(q0 + q) * qrow_stride + d0;
-      const int off0 = SUBT_OFF(q, d0, DBLK);
-      const int off1 = SUBT_OFF(q, d0 + 8, DBLK);
-      *(f32x4*)(imgq + off0) = *(const f32x4*)(qbase + g);
-      *(f32x4*)(imgq + off1) = *(const f32x4*)(qbase + g + 8);
-      *(f32x4*)(imgdo + off0) = *(const f32x4*)(dobase + g);
-      *(f32x4*)(imgdo + off1) = *(const f32x4*)(dobase + g + 8);
-    }
-    __syncthreads();
+    if (q0 + 32 < S) stage(cur ^ 1, q0 + 32);
 
     // S^T (A=K lds, B=imgq) and dP^T (A=V regs, B=imgdo)
     f32x16 accS = (f32x16)(0.f), accDP = (f32x16)(0.f);
@@ -543,8 +555,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       const bf16x8v ka =
           *(const bf16x8v*)(my_k + ((col * (D * 2) + inrow) ^ ((col & KSWZ) << 4)));
       const int boff = SUBT_OFF(col, c * 16 + hb * 8, DBLK);
-      const bf16x8v qbf = *(const bf16x8v*)(imgq + boff);
-      const bf16x8v dbf = *(const bf16x8v*)(imgdo + boff);
+      const bf16x8v qbf = *(const bf16x8v*)(QIMG(cur) + boff);
+      const bf16x8v dbf = *(const bf16x8v*)(DOIMG(cur) + boff);
       accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qbf, accS, 0, 0, 0);
       accDP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vreg[c], dbf, accDP,
                                                       0, 0, 0);
@@ -584,7 +596,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
           const int qg0 = kc * 16 + hb * 8;
           const int rg0 = qg0 >> 2, rg1 = (qg0 + 4) >> 2;
           const bf16x8v bb = tr_read2(
-              imgdo,
+              DOIMG(cur),
               (dvblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
               (dvblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
           accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kc], bb,
@@ -613,7 +625,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
           const int qg0 = kc * 16 + hb * 8;
           const int rg0 = qg0 >> 2, rg1 = (qg0 + 4) >> 2;
           const bf16x8v bb = tr_read2(
-              imgq,
+              QIMG(cur),
               (dkblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
               (dkblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
           accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[kc], bb,
@@ -622,7 +634,10 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       }
     }
     __syncthreads();
+    cur ^= 1;
   }
+#undef QIMG
+#undef DOIMG
 
   // write dK/dV. D-layout: row i = key_rel = DROW(r,hb), col j = feature
   // = t*32 + (lane&31). GQA head-groups collide on (b, key, kvh) ->
@@ -674,20 +689,20 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
       (const short*)do_, (const short*)o, delta_ws, D, S, H, rows);
   dim3 grid(S / 128, B * H);
   if (D == 128) {
-    const int lds_dq = 2 * 32 * 128 * 2;   // subtiled k + v images
+    const int lds_dq = 4 * 32 * 128 * 2;   // dbuf subtiled k+v images
     attn_bwd_dq_kernel<128><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
-    const int lds_dkv = 4 * 32 * 128 * 2 + 2 * 32 * 128 * 2 + 4 * 32 * 64;
+    const int lds_dkv = 4 * 32 * 128 * 2 + 4 * 32 * 128 * 2 + 4 * 32 * 64;
     attn_bwd_dkv_kernel<128><<<grid, 256, lds_dkv, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (float*)dk, (float*)dv, B, S, H, KVH, scale);
   } else {
-    const int lds_dq = 2 * 32 * 64 * 2;
+    const int lds_dq = 4 * 32 * 64 * 2;
     attn_bwd_dq_kernel<64><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
-    const int lds_dkv = 4 * 32 * 64 * 2 + 2 * 32 * 64 * 2 + 4 * 32 * 64;
+    const int lds_dkv = 4 * 32 * 64 * 2 + 4 * 32 * 64 * 2 + 4 * 32 * 64;
     attn_bwd_dkv_kernel<64><<<grid, 256, lds_dkv, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (float*)dk, (float*)dv, B, S, H, KVH, scale);
