@@ -30,12 +30,12 @@ __device__ __forceinline__ float bf2f(short u) {
 }
 
 __device__ __forceinline__ short f2bf(float f) {
-  // round-to-nearest-even bf16
-  union { float f; unsigned int i; } c;
-  c.f = f;
-  unsigned int lsb = (c.i >> 16) & 1;
-  c.i += 0x7fff + lsb;
-  return (short)(c.i >> 16);
+  // compiler-native RNE conversion: pairs fuse into v_cvt_pk_bf16_f32
+  // (manual bit-math RNE costs ~4 VALU ops per element)
+  __bf16 h = (__bf16)f;
+  union { __bf16 b; short s; } c;
+  c.b = h;
+  return c.s;
 }
 
 // full-wave (64-lane) sum reduction
