@@ -120,7 +120,7 @@ class Attention(nn.Module):
         self.qkv = nn.Linear(cfg.emb_dim, qkv_out, bias=False)
         self.proj = nn.Linear(cfg.emb_dim, cfg.emb_dim, bias=False)
 
-    def forward(self, x, cos, sin, cache=None):
+    def forward(self, x, cos, sin, cache=None, residual=None):
         b, s, _ = x.shape
         qkv = self.qkv(x)
         q, k, v = qkv.split(
@@ -145,7 +145,14 @@ class Attention(nn.Module):
                 enable_gqa=(self.kvheads != self.nheads)).transpose(1, 2)
         else:
             o = ops.attention_causal(q, k, v)      # (b, s, nheads, head_dim)
-        return self.proj(o.reshape(b, s, -1))
+        o2 = o.reshape(b * s, -1)
+        if residual is not None and \
+                not getattr(self, "_disable_fused_residual", False):
+            # residual add fused into the output-projection GEMM epilogue
+            return torch.addmm(residual.reshape(b * s, -1), o2,
+                               self.proj.weight.t()).view(b, s, -1)
+        out = self.proj(o2).view(b, s, -1)
+        return out if residual is None else out + residual
 
     def reset_parameters(self):
         for lin in (self.qkv, self.proj):
@@ -159,10 +166,17 @@ class SwiGLU(nn.Module):
         self.wg1 = nn.Linear(cfg.emb_dim, 2 * cfg.hidden_dim, bias=False)  # fused gate|up
         self.w2 = nn.Linear(cfg.hidden_dim, cfg.emb_dim, bias=False)
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         gu = self.wg1(x)
         h = ops.swiglu(gu)                         # silu(g) * u, fused kernel
-        return self.w2(h)
+        if residual is not None and \
+                not getattr(self, "_disable_fused_residual", False):
+            b, s, e = residual.shape
+            return torch.addmm(residual.reshape(b * s, e),
+                               h.reshape(b * s, -1),
+                               self.w2.weight.t()).view(b, s, e)
+        out = self.w2(h)
+        return out if residual is None else out + residual.view_as(out)
 
     def reset_parameters(self):
         for lin in (self.wg1, self.w2):
@@ -188,10 +202,9 @@ class LlamaBlock(nn.Module):
         return self._forward_impl(x, cos, sin, cache)
 
     def _forward_impl(self, x, cos, sin, cache=None):
-        h = self.attn(self.attn_norm(x), cos, sin, cache)
-        # fused residual-add + norm: y2 = rmsnorm(x + h), s = x + h
-        y2, s = ops.add_rmsnorm(x, h, self.mlp_norm.weight, self.mlp_norm.eps)
-        return s + self.mlp(y2)
+        # both residual adds ride the projection GEMM epilogues (addmm)
+        h = self.attn(self.attn_norm(x), cos, sin, cache, residual=x)
+        return self.mlp(self.mlp_norm(h), residual=h)
 
     def reset_parameters(self):
         for m in (self.attn_norm, self.attn, self.mlp_norm, self.mlp):

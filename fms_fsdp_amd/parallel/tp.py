@@ -55,6 +55,7 @@ def tp_shard_llama(model: Llama, tp_group) -> Llama:
         attn.proj.weight.data = _slice_cols(attn.proj.weight.data, rank, ws)
         attn.nheads //= ws
         attn.kvheads //= ws
+        attn._disable_fused_residual = True  # all-reduce rides the module hook
         attn.proj.register_forward_hook(reduce_hook)
 
         mlp: SwiGLU = block.mlp
@@ -64,5 +65,6 @@ def tp_shard_llama(model: Llama, tp_group) -> Llama:
             [_slice_rows(gw, rank, ws), _slice_rows(uw, rank, ws)], dim=0)
         mlp.w2.weight.data = _slice_cols(mlp.w2.weight.data, rank, ws)
         mlp.hidden_dim //= ws
+        mlp._disable_fused_residual = True
         mlp.w2.register_forward_hook(reduce_hook)
     return model
