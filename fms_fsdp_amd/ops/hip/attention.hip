@@ -459,12 +459,12 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 // B-operands). The P/dS lane<->reg transpose goes through a small
 // wave-private LDS buffer. TWO barriers per q-iteration.
 // ---------------------------------------------------------------------
-template <int D>
+template <int D, bool OUT_BF16>
 __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const short* __restrict__ dog, const short* __restrict__ qg,
     const short* __restrict__ kg, const short* __restrict__ vg,
     const float* __restrict__ lseg, const float* __restrict__ deltag,
-    float* __restrict__ dkg, float* __restrict__ dvg, int B, int S, int H,
+    void* __restrict__ dkg, void* __restrict__ dvg, int B, int S, int H,
     int KVH, float scale) {
   constexpr int KVB = 32;   // keys per wave; block = 4 waves = 128 keys
   constexpr int KSWZ = (D == 128) ? 15 : 7;
@@ -640,24 +640,27 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #undef DOIMG
 
   // write dK/dV. D-layout: row i = key_rel = DROW(r,hb), col j = feature
-  // = t*32 + (lane&31). GQA head-groups collide on (b, key, kvh) ->
-  // atomicAdd fp32 when ngrp>1, plain add otherwise (buffers zeroed).
+  // = t*32 + (lane&31). OUT_BF16 (no GQA): each element written exactly
+  // once -> direct bf16 stores, no fp32 scratch/memset/convert pass.
+  // GQA (ngrp>1): head-groups collide on (b, key, kvh) -> fp32 atomics.
 #pragma unroll
   for (int t = 0; t < NT; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key_abs = kv0 + DROW(r, hb);
       const int feat = t * 32 + col;
-      float* dkp = dkg + ((long long)b * S + key_abs) * (long long)(KVH * D) +
-                   (long long)kvh * D;
-      float* dvp = dvg + ((long long)b * S + key_abs) * (long long)(KVH * D) +
-                   (long long)kvh * D;
-      if (ngrp > 1) {
-        atomicAdd(&dkp[feat], accDK[t][r]);
-        atomicAdd(&dvp[feat], accDV[t][r]);
+      const long long off =
+          ((long long)b * S + key_abs) * (long long)(KVH * D) +
+          (long long)kvh * D + feat;
+      if (OUT_BF16) {
+        ((short*)dkg)[off] = f2bf(accDK[t][r]);
+        ((short*)dvg)[off] = f2bf(accDV[t][r]);
+      } else if (ngrp > 1) {
+        atomicAdd((float*)dkg + off, accDK[t][r]);
+        atomicAdd((float*)dvg + off, accDV[t][r]);
       } else {
-        dkp[feat] += accDK[t][r];
-        dvp[feat] += accDV[t][r];
+        ((float*)dkg)[off] += accDK[t][r];
+        ((float*)dvg)[off] += accDV[t][r];
       }
     }
   }
@@ -683,7 +686,8 @@ void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
 void launch_attn_bwd(const void* do_, const void* q, const void* k,
                      const void* v, const void* o, const float* lse, void* dq,
                      void* dk, void* dv, float* delta_ws, int B, int S, int H,
-                     int KVH, int D, float scale, hipStream_t stream) {
+                     int KVH, int D, float scale, int out_bf16,
+                     hipStream_t stream) {
   const long long rows = (long long)B * S * H;
   attn_bwd_delta_kernel<<<(int)((rows * 64 + 255) / 256), 256, 0, stream>>>(
       (const short*)do_, (const short*)o, delta_ws, D, S, H, rows);
@@ -694,18 +698,28 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
     const int lds_dkv = 4 * 32 * 128 * 2 + 4 * 32 * 128 * 2 + 4 * 32 * 64;
-    attn_bwd_dkv_kernel<128><<<grid, 256, lds_dkv, stream>>>(
-        (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
-        lse, delta_ws, (float*)dk, (float*)dv, B, S, H, KVH, scale);
+    if (out_bf16)
+      attn_bwd_dkv_kernel<128, true><<<grid, 256, lds_dkv, stream>>>(
+          (const short*)do_, (const short*)q, (const short*)k,
+          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale);
+    else
+      attn_bwd_dkv_kernel<128, false><<<grid, 256, lds_dkv, stream>>>(
+          (const short*)do_, (const short*)q, (const short*)k,
+          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale);
   } else {
     const int lds_dq = 4 * 32 * 64 * 2;
     attn_bwd_dq_kernel<64><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
     const int lds_dkv = 4 * 32 * 64 * 2 + 4 * 32 * 64 * 2 + 4 * 32 * 64;
-    attn_bwd_dkv_kernel<64><<<grid, 256, lds_dkv, stream>>>(
-        (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
-        lse, delta_ws, (float*)dk, (float*)dv, B, S, H, KVH, scale);
+    if (out_bf16)
+      attn_bwd_dkv_kernel<64, true><<<grid, 256, lds_dkv, stream>>>(
+          (const short*)do_, (const short*)q, (const short*)k,
+          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale);
+    else
+      attn_bwd_dkv_kernel<64, false><<<grid, 256, lds_dkv, stream>>>(
+          (const short*)do_, (const short*)q, (const short*)k,
+          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale);
   }
 }
 

@@ -30,7 +30,7 @@ void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, float, hipStream_t);
 void launch_attn_bwd(const void*, const void*, const void*, const void*,
                      const void*, const float*, void*, void*, void*, float*,
-                     int, int, int, int, int, float, hipStream_t);
+                     int, int, int, int, int, float, int, hipStream_t);
 void launch_cconv_fwd(const void*, const void*, const float*, void*, int,
                       int, int, int, hipStream_t);
 void launch_cconv_bwd(const void*, const void*, const void*, const float*,
@@ -186,15 +186,19 @@ std::tuple<Tensor, Tensor, Tensor> attn_bwd(Tensor do_, Tensor q, Tensor k,
   const int b = q.size(0), s = q.size(1), h = q.size(2), d = q.size(3);
   const int kvh = k.size(2);
   auto dq = torch::empty_like(q);
-  // dk/dv accumulated in fp32 (GQA head-groups collide), cast after
-  auto dk = torch::zeros({b, s, kvh, d}, q.options().dtype(torch::kFloat32));
-  auto dv = torch::zeros({b, s, kvh, d}, q.options().dtype(torch::kFloat32));
+  const bool out_bf16 = (kvh == h);  // no GQA collisions: direct bf16
+  auto opts = q.options().dtype(out_bf16 ? torch::kBFloat16 : torch::kFloat32);
+  auto dk = out_bf16 ? torch::empty({b, s, kvh, d}, opts)
+                     : torch::zeros({b, s, kvh, d}, opts);
+  auto dv = out_bf16 ? torch::empty({b, s, kvh, d}, opts)
+                     : torch::zeros({b, s, kvh, d}, opts);
   auto delta = torch::empty({b, h, s}, q.options().dtype(torch::kFloat32));
   const float scale = 1.f / sqrtf((float)d);
   launch_attn_bwd(do_.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                   o.data_ptr(), lse.data_ptr<float>(), dq.data_ptr(),
                   dk.data_ptr(), dv.data_ptr(), delta.data_ptr<float>(), b, s,
-                  h, kvh, d, scale, cur_stream());
+                  h, kvh, d, scale, out_bf16 ? 1 : 0, cur_stream());
+  if (out_bf16) return {dq, dk, dv};
   return {dq, dk.to(torch::kBFloat16), dv.to(torch::kBFloat16)};
 }
 
