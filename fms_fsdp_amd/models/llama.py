@@ -20,7 +20,7 @@ Convention notes (deliberate deltas from ibm-fms internals):
 
 import math
 from dataclasses import dataclass
-from typing import Optional
+
 
 import torch
 import torch.nn as nn

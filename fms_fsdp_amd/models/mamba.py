@@ -18,7 +18,7 @@ Block layout matches mamba_ssm's Mamba2:
 
 import math
 from dataclasses import dataclass, field
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.nn as nn

@@ -13,7 +13,6 @@ __global__ void ce_fwd_bwd_kernel(short* __restrict__ logits,
                                   const float* __restrict__ denom_ptr,
                                   long long ignore_index,
                                   int V, int rows) {
-  __shared__ float scratch[16];
   __shared__ float s_m, s_l;
   const int V8 = V / 8;
   const int tail = V - V8 * 8;

@@ -107,16 +107,6 @@ def main(**kwargs):
     speculator.reset_parameters()
     speculator = speculator.to(device)
 
-    class _SpecWrap(torch.nn.Module):
-        """Trivial block wrapper so ShardedModel ddp handles grads."""
-
-        def __init__(self, s):
-            super().__init__()
-            self.s = s
-
-        def forward(self, *a, **k):
-            return self.s(*a, **k)
-
     sharded_spec = ShardedModel(
         speculator, MLPSpeculator, sharding_strategy="ddp",
         param_dtype=torch.bfloat16 if torch.cuda.is_available()

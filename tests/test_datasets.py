@@ -123,19 +123,29 @@ def test_bos_injection(corpus):
 
 
 def test_min_length_skips(corpus):
-    d = base_loader(corpus, 0, 1, dataset="dataset_2", min_length=200)
+    """min_length filters short docs: with the threshold above every doc
+    length, stats never advance; below it, everything streams."""
+    d = base_loader(corpus, 0, 1, dataset="dataset_2", min_length=50)
     d.setup()
-    it = iter(d)
-    import itertools
-    got = list(itertools.islice(it, 0))  # all docs are 101 < 200: nothing
-    # iterating one epoch yields nothing; guard with a sentinel loop
-    count = 0
-    gen = iter(d)
-    # one epoch scan: docset_index wraps; just verify no yield in first pass
-    # by checking the generator doesn't produce within a bounded scan
-    # (reach into internals: _len docs, all skipped)
-    assert d._len == 100
-    assert not got and count == 0
+    chunks = take(iter(d), 100)           # all 100 docs pass the filter
+    assert sorted(c[0] // 100 for c in chunks) == list(range(100))
+    d2 = base_loader(corpus, 0, 1, dataset="dataset_2", min_length=102)
+    d2.setup()
+    # docs are 101 tokens incl. delimiter: 101 < 102 -> all skipped; the
+    # epoch scan advances docset_index without yielding
+    it = iter(d2)
+    import threading
+    got = []
+
+    def run():
+        try:
+            got.append(next(it))
+        except StopIteration:
+            pass
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    t.join(timeout=2.0)
+    assert not got, "short docs must be skipped"
 
 
 # ---------------- scalable shards ----------------
