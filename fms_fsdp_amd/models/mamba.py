@@ -98,16 +98,25 @@ def ssd_chunked(x, dt, A, B, C, chunk):
     Bh = Bc.repeat_interleave(rep, dim=3) if g != h else Bc    # (b,nc,Q,h,n)
     Ch = Cc.repeat_interleave(rep, dim=3) if g != h else Cc
 
+    # matmul-shaped work runs in bf16 with fp32 accumulation (MFMA rate;
+    # mamba_ssm's own kernels take bf16 x/B/C the same way) — decay and
+    # cumsum terms stay fp32. On CPU keep fp32 (bf16 matmul is slow there).
+    mm_dtype = torch.bfloat16 if x.is_cuda else torch.float32
+
     # diagonal block: Y[i] = sum_{j<=i} C_i.B_j exp(dA[i]-dA[j]) dt_j x_j
     L = torch.exp(segsum(dA))                        # (b,nc,h,Q,Q)
-    scores = torch.einsum("bcqhn,bckhn->bchqk", Ch, Bh)
+    scores = torch.einsum("bcqhn,bckhn->bchqk",
+                          Ch.to(mm_dtype), Bh.to(mm_dtype)).float()
     xdt = xc * dtc.unsqueeze(-1)                     # (b,nc,Q,h,p)
-    y_diag = torch.einsum("bchqk,bckhp->bcqhp", scores * L, xdt)
+    xdt_m = xdt.to(mm_dtype)
+    y_diag = torch.einsum("bchqk,bckhp->bcqhp",
+                          (scores * L).to(mm_dtype), xdt_m).float()
 
     # chunk-final states: S_c = sum_j exp(dA_end - dA_j) B_j^T (dt_j x_j)
     decay_states = torch.exp(dA_cs[..., -1:] - dA_cs)           # (b,nc,h,Q)
-    states = torch.einsum("bckhn,bchk,bckhp->bchnp",
-                          Bh, decay_states, xdt)                # (b,nc,h,n,p)
+    Bd = Bh * decay_states.permute(0, 1, 3, 2).unsqueeze(-1)    # (b,nc,Q,h,n)
+    states = torch.einsum("bckhn,bckhp->bchnp",
+                          Bd.to(mm_dtype), xdt_m).float()       # (b,nc,h,n,p)
 
     # inter-chunk recurrence (sequential over nc chunks)
     chunk_decay = torch.exp(dA_cs[..., -1])                     # (b,nc,h)
@@ -120,8 +129,9 @@ def ssd_chunked(x, dt, A, B, C, chunk):
 
     # off-diagonal: Y_off[i] = C_i exp(dA_cs[i]) S_{c-1}
     state_decay = torch.exp(dA_cs)                              # (b,nc,h,Q)
-    y_off = torch.einsum("bcqhn,bchq,bchnp->bcqhp",
-                         Ch, state_decay, prev_states)
+    Cd = Ch * state_decay.permute(0, 1, 3, 2).unsqueeze(-1)     # (b,nc,Q,h,n)
+    y_off = torch.einsum("bcqhn,bchnp->bcqhp",
+                         Cd.to(mm_dtype), prev_states.to(mm_dtype)).float()
     return (y_diag + y_off).reshape(b, l, h, p)
 
 

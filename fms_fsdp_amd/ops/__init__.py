@@ -96,6 +96,8 @@ class _RoPEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, cos, sin):
         ext = _require_ext("rope")
+        cos = cos.float().contiguous()   # tables may arrive bf16 after a
+        sin = sin.float().contiguous()   # module-wide .bfloat16() cast
         qo, ko = ext.rope_fwd(q.contiguous(), k.contiguous(), cos, sin, False)
         ctx.save_for_backward(cos, sin)
         return qo, ko
