@@ -49,7 +49,7 @@ class MambaConfig:
     expand: int = 2
     headdim: int = 64
     ngroups: int = 1
-    chunk_size: int = 256
+    chunk_size: int = 128
 
     def __post_init__(self):
         if self.vocab_size % self.pad_vocab_size_multiple != 0:
@@ -104,13 +104,14 @@ def ssd_chunked(x, dt, A, B, C, chunk):
     mm_dtype = torch.bfloat16 if x.is_cuda else torch.float32
 
     # diagonal block: Y[i] = sum_{j<=i} C_i.B_j exp(dA[i]-dA[j]) dt_j x_j
-    L = torch.exp(segsum(dA))                        # (b,nc,h,Q,Q)
+    # the (Q x Q)-sized intermediates dominate HBM traffic: keep the
+    # scores and the masked product in mm_dtype end-to-end
+    L = torch.exp(segsum(dA)).to(mm_dtype)           # (b,nc,h,Q,Q)
     scores = torch.einsum("bcqhn,bckhn->bchqk",
-                          Ch.to(mm_dtype), Bh.to(mm_dtype)).float()
+                          Ch.to(mm_dtype), Bh.to(mm_dtype))
     xdt = xc * dtc.unsqueeze(-1)                     # (b,nc,Q,h,p)
     xdt_m = xdt.to(mm_dtype)
-    y_diag = torch.einsum("bchqk,bckhp->bcqhp",
-                          (scores * L).to(mm_dtype), xdt_m).float()
+    y_diag = torch.einsum("bchqk,bckhp->bcqhp", scores * L, xdt_m).float()
 
     # chunk-final states: S_c = sum_j exp(dA_end - dA_j) B_j^T (dt_j x_j)
     decay_states = torch.exp(dA_cs[..., -1:] - dA_cs)           # (b,nc,h,Q)
