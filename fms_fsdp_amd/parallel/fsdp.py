@@ -156,6 +156,12 @@ class FlatUnit:
         if self.S == 1:
             self._gathered = True
             return
+        if comm_stream is not None:
+            # the shard was last written by the optimizer on the default
+            # stream: the comm stream must order behind that write
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            comm_stream.wait_event(ev)
         ctx = torch.cuda.stream(comm_stream) if comm_stream is not None else nullcontext()
         with ctx:
             self._alloc_flat_param()

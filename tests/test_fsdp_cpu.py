@@ -184,3 +184,18 @@ def test_consolidate_checkpoint(tmp_path):
     sd = consolidate_checkpoint(out)
     for n, t in ref.items():
         assert torch.allclose(sd[n], t.float(), atol=1e-6), n
+
+
+def test_world4_hsdp_two_node_simulation():
+    """world=4 as 2 'nodes' x 2 'GPUs' (intra_node_size=2): exercises the
+    shard-intra + replicate-inter path incl. the inter-node grad
+    all-reduce; must match the single-process trajectory."""
+    ref = _single_process_reference()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_worker, args=(4, PORT + 20, "hsdp", False, q), nprocs=4,
+             join=True)
+    status, got = q.get()
+    assert status == "ok", got
+    for a, b in zip(ref, got):
+        assert abs(a - b) < 1e-5, (ref, got)
