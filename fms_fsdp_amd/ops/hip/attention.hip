@@ -52,6 +52,47 @@ __device__ __forceinline__ bf16x8v pack_pT_chunk(const float* p) {
 #define DROW(r, hb) (((r) & 3) + 8 * ((r) >> 2) + 4 * (hb))
 
 // ---------------------------------------------------------------------
+// Shared helpers for the attention kernels: SUBTILED LDS images.
+// A [R rows][D cols] bf16 tile is stored as [R/4][D/16][4][16] (each
+// subtile 128 B contiguous). This single layout serves BOTH fragment
+// patterns:
+//  - plain B/A reads (row = lane&31, 8 consecutive cols): 16 B contiguous
+//  - transposed reads via gfx950 ds_read_b64_tr_b16: a 16-lane group's
+//    lane j at subtile_base + j*8 receives COLUMN j of the 4x16 subtile
+//    (semantics verified on hardware: tools/tr_probe.hip).
+// ---------------------------------------------------------------------
+// column-block-major subtiles: tile = colblk*R4 + rowgrp (R4 = image
+// rows / 4). A plain 16-lane-group read then touches 4 CONSECUTIVE
+// tiles (alternating LDS bank halves); the rowgrp-bit1 16B XOR spreads
+// the remaining overlap. (PMC-driven: row-group-major was 4-way.)
+#define SUBT_OFF(row, col, R4) \
+  ((((col) >> 4) * (R4) + ((row) >> 2)) * 128 + \
+   ((((row) & 3) * 32 + ((col) & 15) * 2) ^ ((((row) >> 2) & 2) << 3)))
+
+union U2x64 {
+  unsigned long long u[2];
+  bf16x8v v;
+};
+
+// Two transpose reads batched behind ONE lgkmcnt wait (each read's 4
+// shorts = column lane&15 of the 4x16 subtile at its address).
+__device__ __forceinline__ bf16x8v tr_read2(const char* base, int off0,
+                                            int off1) {
+  U2x64 r;
+  const unsigned a0 = (unsigned)(unsigned long long)(base + off0);
+  const unsigned a1 = (unsigned)(unsigned long long)(base + off1);
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(r.u[0]), "=&v"(r.u[1])
+      : "v"(a0), "v"(a1)
+      : "memory");
+  return r.v;
+}
+
+
+// ---------------------------------------------------------------------
 // Forward. Block = 4 waves x 32 q-rows = 128 q rows; KV tile = 64 keys.
 // grid.x = s/128, grid.y = b*h.
 // ---------------------------------------------------------------------
@@ -61,17 +102,16 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const short* __restrict__ vg, short* __restrict__ og,
     float* __restrict__ lseg, int B, int S, int H, int KVH, float scale) {
   constexpr int KVB = 64;
-  constexpr int KSWZ = (D == 128) ? 15 : 7;  // XOR stays inside a D*2-byte row
   constexpr int NC = D / 16;   // QK^T k-chunks
   constexpr int NT = D / 32;   // 32-wide output tiles
+  constexpr int DBLK = D / 16;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // double-buffered: stage tile t+1 while computing tile t, ONE barrier
-  // per tile (the end-of-iter barrier both publishes buf^1 and confirms
-  // every wave is done reading buf before iter t+2 overwrites it)
+  // double-buffered SUBTILED K/V images (same layout as the backward
+  // kernels: coalesced b128 staging, plain reads for the S^T A-operand,
+  // hardware transpose reads for the PV A-operand). ONE barrier/tile.
   constexpr int KB = KVB * D * 2;
-  constexpr int VB = D * KVB * 2;
-#define KLDS(buf) (smem + ((buf) ? (KB + VB) : 0))
-#define VTLDS(buf) (smem + KB + ((buf) ? (KB + VB) : 0))
+#define KLDS(buf) (smem + ((buf) ? 2 * KB : 0))
+#define VLDS(buf) (smem + KB + ((buf) ? 2 * KB : 0))
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -108,34 +148,21 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
   const int t256 = threadIdx.x;
+  // subtiled staging: thread t: key = t&63, colblk pair = t>>6 (each
+  // thread copies 2x16B per image; 64-key rows x D cols)
   auto stage = [&](int buf, int kv0) {
-    // K: 256 threads, each copies KVB*D*2/256 bytes in 16B units
-    constexpr int BYTES_PER_ROW = D * 2;
-    constexpr int CHUNKS = KVB * BYTES_PER_ROW / 16;  // 16B chunks
-#pragma unroll
-    for (int i = t256; i < CHUNKS; i += 256) {
-      const int row = i / (BYTES_PER_ROW / 16);
-      const int cb = (i % (BYTES_PER_ROW / 16)) * 16;
-      const long long g = (long long)(kv0 + row) * krow_stride + cb / 2;
-      *(f32x4*)(KLDS(buf) +
-                ((row * BYTES_PER_ROW + cb) ^ ((row & KSWZ) << 4))) =
-          *(const f32x4*)(kbase + g);
-    }
-    // V transposed: thread t: key = t&63, dv block = (t>>6)*32.
-    // Loads vectorized 16B (guide G13); transposed writes scatter but
-    // stay cheap vs the MFMAs.
     const int key = t256 & 63;
-    const int dv0 = (t256 >> 6) * (D / 4);
-    const short* vp = vbase + (long long)(kv0 + key) * krow_stride + dv0;
+    const int d0 = (t256 >> 6) * (D / 4);   // 32 cols per thread @D=128
 #pragma unroll
-    for (int jj = 0; jj < D / 32; ++jj) {
-      const bf16x8v vv = *(const bf16x8v*)(vp + jj * 8);
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const int dv = dv0 + jj * 8 + e;
-        *(__bf16*)(VTLDS(buf) +
-                   ((dv * (KVB * 2) + key * 2) ^ ((dv & 7) << 4))) = vv[e];
-      }
+    for (int c2 = 0; c2 < D / 64; ++c2) {
+      const int dc = d0 + c2 * 16;
+      const long long g = (long long)(kv0 + key) * krow_stride + dc;
+      const int off0 = SUBT_OFF(key, dc, 16);
+      const int off1 = SUBT_OFF(key, dc + 8, 16);
+      *(f32x4*)(KLDS(buf) + off0) = *(const f32x4*)(kbase + g);
+      *(f32x4*)(KLDS(buf) + off1) = *(const f32x4*)(kbase + g + 8);
+      *(f32x4*)(VLDS(buf) + off0) = *(const f32x4*)(vbase + g);
+      *(f32x4*)(VLDS(buf) + off1) = *(const f32x4*)(vbase + g + 8);
     }
   };
   stage(0, 0);
@@ -160,9 +187,8 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         const int row = kt * 32 + col;
-        const int inrow = c * 32 + hb * 16;
         const bf16x8v a = *(const bf16x8v*)(
-            KLDS(cur) + ((row * (D * 2) + inrow) ^ ((row & KSWZ) << 4)));
+            KLDS(cur) + SUBT_OFF(row, c * 16 + hb * 8, 16));
         accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qb[c], accS, 0, 0, 0);
       }
       float p[16];
@@ -193,22 +219,26 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       }
       l_run = l_run * alpha + s_own + __shfl_xor(s_own, 32, 64);
 
-      // PV for this sub-tile: keys kv32..kv32+31 = chunks 2kt, 2kt+1
+      // PV for this sub-tile: A = V^T via hardware transpose reads on
+      // the row-major V image; keys kv32..kv32+31 = chunks 2kt, 2kt+1
       const bf16x8v pb0 = pack_pT_chunk(p);
       const bf16x8v pb1 = pack_pT_chunk(p + 8);
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
-        const int row = t * 32 + col;   // dv row
-        const int inrow0 = (kt * 2) * 32 + hb * 16;
-        const bf16x8v a0 = *(const bf16x8v*)(
-            VTLDS(cur) + ((row * (KVB * 2) + inrow0) ^ ((row & 7) << 4)));
-        accO[t] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, pb0, accO[t], 0, 0, 0);
-        const int inrow1 = (kt * 2 + 1) * 32 + hb * 16;
-        const bf16x8v a1 = *(const bf16x8v*)(
-            VTLDS(cur) + ((row * (KVB * 2) + inrow1) ^ ((row & 7) << 4)));
-        accO[t] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, pb1, accO[t], 0, 0, 0);
+        const int dvblk = t * 2 + (col >> 4);
+#pragma unroll
+        for (int kc2 = 0; kc2 < 2; ++kc2) {
+          const int key0 = kt * 32 + kc2 * 16 + hb * 8;
+          const int rg0 = key0 >> 2, rg1 = (key0 + 4) >> 2;
+          const bf16x8v a = tr_read2(
+              VLDS(cur),
+              (dvblk * (KVB / 4) + rg0) * 128 +
+                  (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
+              (dvblk * (KVB / 4) + rg1) * 128 +
+                  (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
+          accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a, kc2 ? pb1 : pb0, accO[t], 0, 0, 0);
+        }
       }
     }
     __syncthreads();
@@ -235,7 +265,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     lseg[((long long)bh) * S + my_q] = m_run + __logf(l_run);
 }
 #undef KLDS
-#undef VTLDS
+#undef VLDS
 
 // ---------------------------------------------------------------------
 // Backward preprocess: delta[b,h,s] = rowsum(dO * O) fp32
@@ -261,46 +291,6 @@ __global__ void attn_bwd_delta_kernel(const short* __restrict__ dog,
     const int hi = (int)(row % H);
     delta[(b * H + hi) * (long long)S + si] = acc;
   }
-}
-
-// ---------------------------------------------------------------------
-// Shared helpers for the backward kernels: SUBTILED LDS images.
-// A [R rows][D cols] bf16 tile is stored as [R/4][D/16][4][16] (each
-// subtile 128 B contiguous). This single layout serves BOTH fragment
-// patterns:
-//  - plain B/A reads (row = lane&31, 8 consecutive cols): 16 B contiguous
-//  - transposed reads via gfx950 ds_read_b64_tr_b16: a 16-lane group's
-//    lane j at subtile_base + j*8 receives COLUMN j of the 4x16 subtile
-//    (semantics verified on hardware: tools/tr_probe.hip).
-// ---------------------------------------------------------------------
-// column-block-major subtiles: tile = colblk*(R/4=8) + rowgrp. A plain
-// 16-lane-group read then touches 4 CONSECUTIVE tiles (alternating LDS
-// bank halves); the rowgrp-bit1 16B XOR spreads the remaining overlap.
-// (PMC-driven: the row-group-major variant was a 4-way conflict.)
-#define SUBT_OFF(row, col, DBLK) \
-  ((((col) >> 4) * 8 + ((row) >> 2)) * 128 + \
-   ((((row) & 3) * 32 + ((col) & 15) * 2) ^ ((((row) >> 2) & 2) << 3)))
-
-union U2x64 {
-  unsigned long long u[2];
-  bf16x8v v;
-};
-
-// Two transpose reads batched behind ONE lgkmcnt wait (each read's 4
-// shorts = column lane&15 of the 4x16 subtile at its address).
-__device__ __forceinline__ bf16x8v tr_read2(const char* base, int off0,
-                                            int off1) {
-  U2x64 r;
-  const unsigned a0 = (unsigned)(unsigned long long)(base + off0);
-  const unsigned a1 = (unsigned)(unsigned long long)(base + off1);
-  asm volatile(
-      "ds_read_b64_tr_b16 %0, %2\n\t"
-      "ds_read_b64_tr_b16 %1, %3\n\t"
-      "s_waitcnt lgkmcnt(0)"
-      : "=&v"(r.u[0]), "=&v"(r.u[1])
-      : "v"(a0), "v"(a1)
-      : "memory");
-  return r.v;
 }
 
 // ---------------------------------------------------------------------
@@ -374,8 +364,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const int key = t256 & 31;
     const int d0 = (t256 >> 5) * 16;
     const long long g = (long long)(kv0s + key) * krow_stride + d0;
-    const int off0 = SUBT_OFF(key, d0, DBLK);
-    const int off1 = SUBT_OFF(key, d0 + 8, DBLK);
+    const int off0 = SUBT_OFF(key, d0, 8);
+    const int off1 = SUBT_OFF(key, d0 + 8, 8);
     *(f32x4*)(KIMG(buf) + off0) = *(const f32x4*)(kbase + g);
     *(f32x4*)(KIMG(buf) + off1) = *(const f32x4*)(kbase + g + 8);
     *(f32x4*)(VIMG(buf) + off0) = *(const f32x4*)(vbase + g);
@@ -392,7 +382,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     f32x16 accS = (f32x16)(0.f), accDP = (f32x16)(0.f);
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
-      const int off = SUBT_OFF(col, c * 16 + hb * 8, DBLK);
+      const int off = SUBT_OFF(col, c * 16 + hb * 8, 8);
       const bf16x8v ka = *(const bf16x8v*)(KIMG(cur) + off);
       const bf16x8v va = *(const bf16x8v*)(VIMG(cur) + off);
       accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qb[c], accS, 0, 0, 0);
@@ -534,8 +524,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const int q = t256 & 31;
     const int d0 = (t256 >> 5) * 16;
     const long long g = (long long)(q0s + q) * qrow_stride + d0;
-    const int off0 = SUBT_OFF(q, d0, DBLK);
-    const int off1 = SUBT_OFF(q, d0 + 8, DBLK);
+    const int off0 = SUBT_OFF(q, d0, 8);
+    const int off1 = SUBT_OFF(q, d0 + 8, 8);
     *(f32x4*)(QIMG(buf) + off0) = *(const f32x4*)(qbase + g);
     *(f32x4*)(QIMG(buf) + off1) = *(const f32x4*)(qbase + g + 8);
     *(f32x4*)(DOIMG(buf) + off0) = *(const f32x4*)(dobase + g);
@@ -554,7 +544,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       const int inrow = c * 32 + hb * 16;
       const bf16x8v ka =
           *(const bf16x8v*)(my_k + ((col * (D * 2) + inrow) ^ ((col & KSWZ) << 4)));
-      const int boff = SUBT_OFF(col, c * 16 + hb * 8, DBLK);
+      const int boff = SUBT_OFF(col, c * 16 + hb * 8, 8);
       const bf16x8v qbf = *(const bf16x8v*)(QIMG(cur) + boff);
       const bf16x8v dbf = *(const bf16x8v*)(DOIMG(cur) + boff);
       accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qbf, accS, 0, 0, 0);
