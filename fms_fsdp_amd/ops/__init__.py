@@ -256,19 +256,22 @@ def causal_conv1d(x, weight, bias):
 # Fused AdamW on flat fp32 shards + multi-tensor sq-norm
 # --------------------------------------------------------------------------
 def fused_adamw(p, g, m, v, step, lr, beta1, beta2, eps, weight_decay,
-                grad_scale=None):
+                grad_scale=None, p_bf16_out=None):
     """In-place AdamW on 1-D fp32 tensors. g may be fp32 or bf16;
-    grad_scale (0-dim fp32 tensor) folds grad clipping into the update."""
+    grad_scale (0-dim fp32 tensor) folds grad clipping into the update;
+    p_bf16_out receives the updated bf16 shard in the same pass."""
     if p.is_cuda:
         ext = _require_ext("adamw")
         ext.adamw(p, g.view(-1), m, v, float(step), lr, beta1, beta2, eps,
-                  weight_decay, grad_scale)
-    else:
-        gf = g.float().view(-1)
-        if grad_scale is not None:
-            gf = gf * grad_scale
-        reference.adamw_step(p, gf, m, v, step, lr, beta1, beta2, eps,
-                             weight_decay)
+                  weight_decay, grad_scale,
+                  p_bf16_out.view(-1) if p_bf16_out is not None else None)
+        return True
+    gf = g.float().view(-1)
+    if grad_scale is not None:
+        gf = gf * grad_scale
+    reference.adamw_step(p, gf, m, v, step, lr, beta1, beta2, eps,
+                         weight_decay)
+    return False
 
 
 def sq_norm(tensors):

@@ -11,6 +11,7 @@ __global__ void adamw_kernel(float* __restrict__ p,
                              const void* __restrict__ g_,
                              float* __restrict__ m,
                              float* __restrict__ v,
+                             short* __restrict__ p_bf16_out,
                              long long n4, float lr, float b1, float b2,
                              float eps, float wd, float bc1, float bc2,
                              const float* __restrict__ gscale) {
@@ -42,6 +43,14 @@ __global__ void adamw_kernel(float* __restrict__ p,
   ((f32x4*)p)[i] = pv;
   ((f32x4*)m)[i] = mv;
   ((f32x4*)v)[i] = vv;
+  if (p_bf16_out != nullptr) {
+    // publish the updated bf16 shard in the same pass (saves a separate
+    // master->shard cast sweep: ~40 GB/step on a 7B model)
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o.v[j] = f2bf(pv.v[j]);
+    ((bf16x4*)p_bf16_out)[i] = o;
+  }
 }
 
 template <bool BF16>
@@ -68,21 +77,21 @@ __global__ void sqnorm_kernel(const void* __restrict__ t, float* __restrict__ ou
 extern "C" {
 
 void launch_adamw(float* p, const void* g, int grad_is_bf16, float* m,
-                  float* v, long long n, float lr, float b1, float b2,
-                  float eps, float wd, float bc1, float bc2,
-                  const float* gscale, hipStream_t stream) {
+                  float* v, void* p_bf16_out, long long n, float lr,
+                  float b1, float b2, float eps, float wd, float bc1,
+                  float bc2, const float* gscale, hipStream_t stream) {
   // flat shards are 128-element aligned; n % 4 == 0 guaranteed
   const long long n4 = n / 4;
   const int block = 256;
   const long long grid = (n4 + block - 1) / block;
   if (grad_is_bf16)
-    adamw_kernel<true><<<(int)grid, block, 0, stream>>>(p, g, m, v, n4, lr, b1,
-                                                        b2, eps, wd, bc1, bc2,
-                                                        gscale);
+    adamw_kernel<true><<<(int)grid, block, 0, stream>>>(
+        p, g, m, v, (short*)p_bf16_out, n4, lr, b1, b2, eps, wd, bc1, bc2,
+        gscale);
   else
-    adamw_kernel<false><<<(int)grid, block, 0, stream>>>(p, g, m, v, n4, lr, b1,
-                                                         b2, eps, wd, bc1, bc2,
-                                                         gscale);
+    adamw_kernel<false><<<(int)grid, block, 0, stream>>>(
+        p, g, m, v, (short*)p_bf16_out, n4, lr, b1, b2, eps, wd, bc1, bc2,
+        gscale);
 }
 
 void launch_sqnorm(const void* t, int is_bf16, float* out, long long n,

@@ -22,9 +22,9 @@ void launch_swiglu_bwd(const void*, const void*, void*, long long, int,
                        hipStream_t);
 void launch_ce_fwd_bwd(void*, const long long*, float*, const float*,
                        long long, int, int, hipStream_t);
-void launch_adamw(float*, const void*, int, float*, float*, long long, float,
-                  float, float, float, float, float, float, const float*,
-                  hipStream_t);
+void launch_adamw(float*, const void*, int, float*, float*, void*,
+                  long long, float, float, float, float, float, float, float,
+                  const float*, hipStream_t);
 void launch_sqnorm(const void*, int, float*, long long, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, float, hipStream_t);
@@ -141,7 +141,8 @@ void ce_fwd_bwd(Tensor logits, Tensor labels, Tensor loss_sum, Tensor denom,
 
 void adamw(Tensor p, Tensor g, Tensor m, Tensor v, double step, double lr,
            double b1, double b2, double eps, double wd,
-           c10::optional<Tensor> grad_scale) {
+           c10::optional<Tensor> grad_scale,
+           c10::optional<Tensor> p_bf16_out) {
   TORCH_CHECK(p.scalar_type() == torch::kFloat32 && p.is_contiguous());
   const long long n = p.numel();
   TORCH_CHECK(n % 4 == 0, "shard size must be divisible by 4");
@@ -149,8 +150,10 @@ void adamw(Tensor p, Tensor g, Tensor m, Tensor v, double step, double lr,
   const float bc1 = 1.f - powf((float)b1, (float)step);
   const float bc2 = 1.f - powf((float)b2, (float)step);
   launch_adamw(p.data_ptr<float>(), g.data_ptr(), gbf, m.data_ptr<float>(),
-               v.data_ptr<float>(), n, (float)lr, (float)b1, (float)b2,
-               (float)eps, (float)wd, bc1, bc2,
+               v.data_ptr<float>(),
+               p_bf16_out.has_value() ? p_bf16_out->data_ptr() : nullptr, n,
+               (float)lr, (float)b1, (float)b2, (float)eps, (float)wd, bc1,
+               bc2,
                grad_scale.has_value() ? grad_scale->data_ptr<float>() : nullptr,
                cur_stream());
 }

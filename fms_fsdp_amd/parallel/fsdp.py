@@ -425,10 +425,14 @@ class ShardedAdamW:
         gscale = getattr(self.m, "_clip_coef", None)
         for u in self.m.all_units:
             u.wait_grads()
-            ops.fused_adamw(u.master_shard, u.grad_shard, u.exp_avg,
-                            u.exp_avg_sq, self.step_count, lr, b1, b2,
-                            self.eps, self.weight_decay, grad_scale=gscale)
-            u.publish_master_to_shard()
+            published = ops.fused_adamw(
+                u.master_shard, u.grad_shard, u.exp_avg, u.exp_avg_sq,
+                self.step_count, lr, b1, b2, self.eps, self.weight_decay,
+                grad_scale=gscale,
+                p_bf16_out=u.param_shard
+                if u.param_shard.dtype == torch.bfloat16 else None)
+            if not published or u.param_shard.dtype != torch.bfloat16:
+                u.publish_master_to_shard()
             u.mark_stale()
 
     def zero_grad(self, set_to_none=False):

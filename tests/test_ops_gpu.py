@@ -109,7 +109,7 @@ def test_adamw_matches_torch():
     for step in range(1, 4):
         p_ref.grad = g.clone()
         opt.step()
-        _C.adamw(p, g, m, v, float(step), 1e-3, 0.9, 0.95, 1e-8, 0.1, None)
+        _C.adamw(p, g, m, v, float(step), 1e-3, 0.9, 0.95, 1e-8, 0.1, None, None)
     assert relerr(p, p_ref.detach()) < 1e-4
 
 
