@@ -105,9 +105,11 @@ def ssd_chunked(x, dt, A, B, C, chunk):
     mm_dtype = torch.bfloat16 if x.is_cuda else torch.float32
 
     # diagonal block: Y[i] = sum_{j<=i} C_i.B_j exp(dA[i]-dA[j]) dt_j x_j
-    # the (Q x Q)-sized intermediates dominate HBM traffic: keep the
-    # scores and the masked product in mm_dtype end-to-end
-    L = torch.exp(segsum(dA)).to(mm_dtype)           # (b,nc,h,Q,Q)
+    # the (Q x Q)-sized intermediates dominate HBM traffic: L is emitted
+    # bf16 in ONE fused kernel pass (ops/hip/ssd.hip) and the scores /
+    # masked product stay in mm_dtype end-to-end
+    from fms_fsdp_amd.ops import segsum_exp
+    L = segsum_exp(dA_cs).to(mm_dtype)               # (b,nc,h,Q,Q)
     scores = torch.einsum("bcqhn,bckhn->bchqk",
                           Ch.to(mm_dtype), Bh.to(mm_dtype))
     xdt = xc * dtc.unsqueeze(-1)                     # (b,nc,Q,h,p)

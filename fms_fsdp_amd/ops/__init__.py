@@ -253,6 +253,35 @@ def causal_conv1d(x, weight, bias):
 
 
 # --------------------------------------------------------------------------
+# Fused exp(segsum) for the Mamba2 SSD scan: L[n,i,j] = exp(cs_i - cs_j)
+# lower-triangular, emitted bf16 in one pass.
+# --------------------------------------------------------------------------
+class _SegsumExpFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, cs):
+        ext = _require_ext("segsum_exp")
+        cs = cs.contiguous()
+        L = ext.segsum_exp_fwd(cs)
+        ctx.save_for_backward(cs)
+        return L
+
+    @staticmethod
+    def backward(ctx, gL):
+        (cs,) = ctx.saved_tensors
+        return _C.segsum_exp_bwd(gL.contiguous().to(torch.bfloat16), cs)
+
+
+def segsum_exp(cs):
+    """cs (..., Q) fp32 cumulative sums -> bf16 (..., Q, Q) decay matrix."""
+    if cs.is_cuda:
+        return _SegsumExpFn.apply(cs)
+    Q = cs.shape[-1]
+    out = cs[..., :, None] - cs[..., None, :]
+    mask = torch.tril(torch.ones(Q, Q, dtype=torch.bool, device=cs.device), 0)
+    return torch.exp(out.masked_fill(~mask, -torch.inf))
+
+
+# --------------------------------------------------------------------------
 # Fused AdamW on flat fp32 shards + multi-tensor sq-norm
 # --------------------------------------------------------------------------
 def fused_adamw(p, g, m, v, step, lr, beta1, beta2, eps, weight_decay,

@@ -36,6 +36,9 @@ void launch_cconv_fwd(const void*, const void*, const float*, void*, int,
 void launch_cconv_bwd(const void*, const void*, const void*, const float*,
                       void*, void*, float*, float*, int, int, int, int,
                       hipStream_t);
+void launch_segsum_exp_fwd(const float*, void*, long long, int, hipStream_t);
+void launch_segsum_exp_bwd(const void*, const float*, float*, long long, int,
+                           hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -235,7 +238,32 @@ std::tuple<Tensor, Tensor, Tensor> cconv_bwd(Tensor dy, Tensor x, Tensor w,
   return {dx, dw, db};
 }
 
+Tensor segsum_exp_fwd(Tensor cs) {
+  TORCH_CHECK(cs.scalar_type() == torch::kFloat32 && cs.is_contiguous());
+  const long long N = cs.numel() / cs.size(-1);
+  const int Q = cs.size(-1);
+  TORCH_CHECK(Q % 8 == 0);
+  std::vector<int64_t> shape(cs.sizes().begin(), cs.sizes().end());
+  shape.push_back(Q);
+  auto out = torch::empty(shape, cs.options().dtype(torch::kBFloat16));
+  launch_segsum_exp_fwd(cs.data_ptr<float>(), out.data_ptr(), N, Q,
+                        cur_stream());
+  return out;
+}
+
+Tensor segsum_exp_bwd(Tensor g, Tensor cs) {
+  CHECK_BF16_CONTIG(g);
+  const long long N = cs.numel() / cs.size(-1);
+  const int Q = cs.size(-1);
+  auto dcs = torch::empty_like(cs);
+  launch_segsum_exp_bwd(g.data_ptr(), cs.data_ptr<float>(),
+                        dcs.data_ptr<float>(), N, Q, cur_stream());
+  return dcs;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("segsum_exp_fwd", &segsum_exp_fwd);
+  mod.def("segsum_exp_bwd", &segsum_exp_bwd);
   mod.def("cconv_fwd", &cconv_fwd);
   mod.def("cconv_bwd", &cconv_bwd);
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);

@@ -340,3 +340,22 @@ def test_speculator_stage2_gpu(tmp_path):
     loss, stats, ntok = stage2_loss(cfg, m, spec, inp, inp, stats)
     assert torch.isfinite(loss)
     loss.backward()
+
+
+def test_segsum_exp():
+    torch.manual_seed(9)
+    from fms_fsdp_amd import _C
+    from fms_fsdp_amd.ops import reference  # noqa
+    N, Q = 6, 128
+    cs = (-torch.rand(N, Q, device=dev()).cumsum(-1)).contiguous()
+    L = _C.segsum_exp_fwd(cs)
+    mask = torch.tril(torch.ones(Q, Q, dtype=torch.bool, device=dev()), 0)
+    ref = torch.exp((cs[:, :, None] - cs[:, None, :]).masked_fill(~mask, -torch.inf))
+    assert relerr(L, ref) < 1e-2
+    # backward vs autograd
+    csf = cs.clone().requires_grad_()
+    Lf = torch.exp((csf[:, :, None] - csf[:, None, :]).masked_fill(~mask, -torch.inf))
+    g = torch.randn_like(Lf)
+    Lf.backward(g)
+    dcs = _C.segsum_exp_bwd(g.bfloat16().contiguous(), cs)
+    assert relerr(dcs, csf.grad) < 5e-2
