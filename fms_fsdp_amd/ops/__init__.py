@@ -184,7 +184,9 @@ class _LinearCEFn(torch.autograd.Function):
     would be ~8 GB fp32 at b2 s8192).
     """
 
-    CHUNK = 2048  # rows per chunk; 2048 x 128256 bf16 logits = 525 MB -> L2/HBM friendly
+    # rows per chunk: 8192 x 128256 bf16 logits = 2.1 GB transient — easily
+    # afforded by 288 GB HBM, and the lm-head GEMMs run at full M
+    CHUNK = 8192
 
     @staticmethod
     def forward(ctx, x, weight, labels, ignore_index):
