@@ -119,18 +119,40 @@ def rope_apply(q, k, cos, sin):
 # Causal flash attention (MFMA tiled, GQA)
 # --------------------------------------------------------------------------
 class _FlashAttnFn(torch.autograd.Function):
+    """Kernel operates on seq multiples of 128; arbitrary lengths are
+    zero-padded at the END (causality keeps pad keys invisible to real
+    queries) and sliced back."""
+
+    @staticmethod
+    def _pad(t, s_pad):
+        b, s, h, d = t.shape
+        if s == s_pad:
+            return t.contiguous()
+        out = torch.zeros(b, s_pad, h, d, dtype=t.dtype, device=t.device)
+        out[:, :s] = t
+        return out
+
     @staticmethod
     def forward(ctx, q, k, v):
         ext = _require_ext("attention")
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        o, lse = ext.attn_fwd(q, k, v)
-        ctx.save_for_backward(q, k, v, o, lse)
-        return o
+        s = q.shape[1]
+        s_pad = (s + 127) // 128 * 128
+        qp = _FlashAttnFn._pad(q, s_pad)
+        kp = _FlashAttnFn._pad(k, s_pad)
+        vp = _FlashAttnFn._pad(v, s_pad)
+        o, lse = ext.attn_fwd(qp, kp, vp)
+        ctx.save_for_backward(qp, kp, vp, o, lse)
+        ctx.s = s
+        return o[:, :s] if s != s_pad else o
 
     @staticmethod
     def backward(ctx, do):
-        q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = _C.attn_bwd(do.contiguous(), q, k, v, o, lse)
+        qp, kp, vp, o, lse = ctx.saved_tensors
+        s, s_pad = ctx.s, qp.shape[1]
+        dop = _FlashAttnFn._pad(do, s_pad)
+        dq, dk, dv = _C.attn_bwd(dop, qp, kp, vp, o, lse)
+        if s != s_pad:
+            dq, dk, dv = dq[:, :s], dk[:, :s], dv[:, :s]
         return dq, dk, dv
 
 

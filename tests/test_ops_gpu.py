@@ -359,3 +359,21 @@ def test_segsum_exp():
     Lf.backward(g)
     dcs = _C.segsum_exp_bwd(g.bfloat16().contiguous(), cs)
     assert relerr(dcs, csf.grad) < 5e-2
+
+
+def test_attention_padded_seq():
+    """Arbitrary (non-128-multiple) seq lens via end-padding dispatch."""
+    from fms_fsdp_amd import ops
+    torch.manual_seed(11)
+    b, s, h, kvh, d = 2, 200, 2, 2, 128
+    q = torch.randn(b, s, h, d, device=dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(b, s, kvh, d, device=dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(b, s, kvh, d, device=dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = ops.attention_causal(q, k, v)
+    ref = reference.attention_causal(q.detach(), k.detach(), v.detach())
+    assert relerr(o, ref) < 3e-2
+    o.sum().backward()
+    assert torch.isfinite(q.grad).all() and torch.isfinite(k.grad).all()
