@@ -329,6 +329,7 @@ def fused_adamw(p, g, m, v, step, lr, beta1, beta2, eps, weight_decay,
 
 def sq_norm(tensors):
     """Sum of squares over a list of tensors -> fp32 scalar tensor."""
+    tensors = [t for t in tensors if t.numel() > 0]
     if tensors and tensors[0].is_cuda and _C is not None:
         out = torch.zeros((), device=tensors[0].device, dtype=torch.float32)
         for t in tensors:

@@ -521,15 +521,16 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   const int q_start = (blockIdx.x * 128) / 32 * 32;
   const int t256 = threadIdx.x;
   auto stage = [&](int buf, int q0s) {
-    const int q = t256 & 31;
-    const int d0 = (t256 >> 5) * 16;
-    const long long g = (long long)(q0s + q) * qrow_stride + d0;
-    const int off0 = SUBT_OFF(q, d0, 8);
-    const int off1 = SUBT_OFF(q, d0 + 8, 8);
-    *(f32x4*)(QIMG(buf) + off0) = *(const f32x4*)(qbase + g);
-    *(f32x4*)(QIMG(buf) + off1) = *(const f32x4*)(qbase + g + 8);
-    *(f32x4*)(DOIMG(buf) + off0) = *(const f32x4*)(dobase + g);
-    *(f32x4*)(DOIMG(buf) + off1) = *(const f32x4*)(dobase + g + 8);
+    constexpr int CPR = D / 8;            // 16B chunks per row
+#pragma unroll
+    for (int i = t256; i < 32 * CPR; i += 256) {
+      const int q = i % 32;
+      const int d0 = (i / 32) * 8;
+      const long long g = (long long)(q0s + q) * qrow_stride + d0;
+      const int off = SUBT_OFF(q, d0, 8);
+      *(f32x4*)(QIMG(buf) + off) = *(const f32x4*)(qbase + g);
+      *(f32x4*)(DOIMG(buf) + off) = *(const f32x4*)(dobase + g);
+    }
   };
   stage(0, q_start);
   __syncthreads();

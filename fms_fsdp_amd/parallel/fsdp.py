@@ -290,10 +290,14 @@ class ShardedModel(nn.Module):
                                        param_names=[name_of[id(p)] for p in params]))
         root_params = [p for p in model.parameters()
                        if p.requires_grad and id(p) not in block_param_ids]
-        self.root_unit = FlatUnit("_root", model, root_params, shard_group,
-                                  replicate_group, device, param_dtype, False,
-                                  param_names=[name_of[id(p)] for p in root_params])
-        self.all_units = [self.root_unit] + self.units
+        self.root_unit = None
+        if root_params:
+            self.root_unit = FlatUnit(
+                "_root", model, root_params, shard_group, replicate_group,
+                device, param_dtype, False,
+                param_names=[name_of[id(p)] for p in root_params])
+        self.all_units = ([self.root_unit] if self.root_unit else []) \
+            + self.units
         self._unit_of_module = {id(u.module): u for u in self.units}
 
         # move buffers (e.g. rope tables) to device, keep dtype
@@ -320,6 +324,7 @@ class ShardedModel(nn.Module):
         for u in self.all_units:
             for p in u.params:
                 p.register_post_accumulate_grad_hook(self._make_grad_hook(u))
+        # no separate root unit: nothing extra to gather at forward start
 
     def _make_fwd_pre(self, idx):
         def hook(module, args):
@@ -361,14 +366,15 @@ class ShardedModel(nn.Module):
             u._grad_countdown -= 1
             if u._grad_countdown == 0:
                 u.reduce_grads(self.rs_stream, self.reduce_dtype)
-                if u is not self.root_unit and u.reshard_after_forward:
+                if u.reshard_after_forward:
                     u._free_flat_param()
         return hook
 
     def forward(self, *args, **kwargs):
         # root params (embedding) needed first; gather root + lookahead
-        self.root_unit.gather(self.comm_stream)
-        self.root_unit.wait_gather()
+        if self.root_unit is not None:
+            self.root_unit.gather(self.comm_stream)
+            self.root_unit.wait_gather()
         for j in range(min(self.prefetch_lookahead, len(self.units))):
             self.units[j].gather(self.comm_stream)
         return self.model(*args, **kwargs)

@@ -150,6 +150,8 @@ def test_attention_fwd(b, s, h, kvh, d):
 @pytest.mark.parametrize("b,s,h,kvh,d", [
     (1, 128, 2, 2, 128),
     (2, 256, 4, 2, 128),
+    (1, 256, 4, 4, 64),
+    (2, 1024, 12, 12, 64),
 ])
 def test_attention_bwd(b, s, h, kvh, d):
     torch.manual_seed(6)
