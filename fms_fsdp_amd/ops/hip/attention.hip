@@ -358,18 +358,19 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
   const int t256 = threadIdx.x;
-  // stage K/V into subtiled images: thread t: key = t&31, colblk = t>>5
-  // (8-lane write groups span adjacent rowgrp tiles -> conflict-free)
+  // stage K/V into subtiled images (generic over D: 32*D/8 16B chunks;
+  // 8-lane write groups span adjacent rowgrp tiles -> conflict-free)
   auto stage = [&](int buf, int kv0s) {
-    const int key = t256 & 31;
-    const int d0 = (t256 >> 5) * 16;
-    const long long g = (long long)(kv0s + key) * krow_stride + d0;
-    const int off0 = SUBT_OFF(key, d0, 8);
-    const int off1 = SUBT_OFF(key, d0 + 8, 8);
-    *(f32x4*)(KIMG(buf) + off0) = *(const f32x4*)(kbase + g);
-    *(f32x4*)(KIMG(buf) + off1) = *(const f32x4*)(kbase + g + 8);
-    *(f32x4*)(VIMG(buf) + off0) = *(const f32x4*)(vbase + g);
-    *(f32x4*)(VIMG(buf) + off1) = *(const f32x4*)(vbase + g + 8);
+    constexpr int CPR = D / 8;            // 16B chunks per key row
+#pragma unroll
+    for (int i = t256; i < 32 * CPR; i += 256) {
+      const int key = i % 32;
+      const int d0 = (i / 32) * 8;
+      const long long g = (long long)(kv0s + key) * krow_stride + d0;
+      const int off = SUBT_OFF(key, d0, 8);
+      *(f32x4*)(KIMG(buf) + off) = *(const f32x4*)(kbase + g);
+      *(f32x4*)(VIMG(buf) + off) = *(const f32x4*)(vbase + g);
+    }
   };
   stage(0, 0);
   __syncthreads();
