@@ -19,6 +19,7 @@ Convention notes (deliberate deltas from ibm-fms internals):
 """
 
 import math
+import os
 from dataclasses import dataclass
 
 
@@ -123,6 +124,23 @@ class Attention(nn.Module):
     def forward(self, x, cos, sin, cache=None, residual=None):
         b, s, _ = x.shape
         qkv = self.qkv(x)
+        if (cache is None and qkv.is_cuda and s % 128 == 0
+                and self.head_dim in (64, 128)
+                and qkv.dtype == torch.bfloat16
+                and os.environ.get("FMS_AMD_ALLOW_TORCH_SDPA") != "1"
+                and os.environ.get("FMS_AMD_DISABLE_FUSED_QKV") != "1"):
+            # fused split+RoPE+attention: strided kernels read q/k/v
+            # straight from the fused projection; backward fills one
+            # fused dqkv buffer (no cat/copies)
+            o = ops.qkv_rope_attention(qkv, cos, sin, self.nheads,
+                                       self.kvheads, self.head_dim)
+            o2 = o.reshape(b * s, -1)
+            if residual is not None and \
+                    not getattr(self, "_disable_fused_residual", False):
+                return torch.addmm(residual.reshape(b * s, -1), o2,
+                                   self.proj.weight.t()).view(b, s, -1)
+            out = self.proj(o2).view(b, s, -1)
+            return out if residual is None else out + residual
         q, k, v = qkv.split(
             [self.nheads * self.head_dim,
              self.kvheads * self.head_dim,

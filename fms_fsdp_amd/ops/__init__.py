@@ -92,20 +92,36 @@ def add_rmsnorm(x, res, weight, eps=1e-6):
 # --------------------------------------------------------------------------
 # RoPE (half-rotation convention; cos/sin tables (s, d/2) fp32)
 # --------------------------------------------------------------------------
+def _row_view_ok(t):
+    """(b, s, h, d) bf16 view the strided kernels accept directly:
+    contiguous within each (b, s) row, uniform row stride (a last-dim
+    slice of a fused qkv projection qualifies — no copy needed)."""
+    return (t.dim() == 4 and t.stride(3) == 1 and t.stride(2) == t.shape[3]
+            and t.stride(1) >= t.shape[2] * t.shape[3]
+            and t.stride(0) == t.shape[1] * t.stride(1))
+
+
+def _as_row_view(t):
+    return t if _row_view_ok(t) else t.contiguous()
+
+
 class _RoPEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, cos, sin):
         ext = _require_ext("rope")
         cos = cos.float().contiguous()   # tables may arrive bf16 after a
         sin = sin.float().contiguous()   # module-wide .bfloat16() cast
-        qo, ko = ext.rope_fwd(q.contiguous(), k.contiguous(), cos, sin, False)
+        # strided reads handle fused-qkv slices without a copy
+        qo, ko = ext.rope_fwd(_as_row_view(q), _as_row_view(k), cos, sin,
+                              False)
         ctx.save_for_backward(cos, sin)
         return qo, ko
 
     @staticmethod
     def backward(ctx, dq, dk):
         cos, sin = ctx.saved_tensors
-        dqo, dko = _C.rope_fwd(dq.contiguous(), dk.contiguous(), cos, sin, True)
+        dqo, dko = _C.rope_fwd(_as_row_view(dq), _as_row_view(dk), cos, sin,
+                               True)
         return dqo, dko, None, None
 
 
@@ -150,10 +166,72 @@ class _FlashAttnFn(torch.autograd.Function):
         qp, kp, vp, o, lse = ctx.saved_tensors
         s, s_pad = ctx.s, qp.shape[1]
         dop = _FlashAttnFn._pad(do, s_pad)
-        dq, dk, dv = _C.attn_bwd(dop, qp, kp, vp, o, lse)
+        dq, dk, dv = _C.attn_bwd(dop, qp, kp, vp, o, lse, None)
         if s != s_pad:
             dq, dk, dv = dq[:, :s], dk[:, :s], dv[:, :s]
         return dq, dk, dv
+
+
+class _QKVRopeAttnFn(torch.autograd.Function):
+    """Fused qkv-split + RoPE + causal flash attention over the FUSED qkv
+    projection (b, s, (h+2kv)*d).
+
+    Removes the per-layer copy traffic the modular path pays: q/k are
+    rotated straight out of their strided qkv slices (strided rope
+    reads), v feeds the attention kernels as a strided view (vstride
+    plumbed into the HIP kernels), and the backward writes dq/dk/dv into
+    ONE preallocated fused dqkv buffer — no autograd `cat` of the split,
+    no zero-fill, no `.contiguous()` copies. Replaces the reference's
+    separate rope+SDPA calls (fms MultiHeadAttention).
+    """
+
+    @staticmethod
+    def forward(ctx, qkv, cos, sin, nheads, kvheads, head_dim):
+        ext = _require_ext("attention")
+        b, s, _ = qkv.shape
+        d = head_dim
+        hq, hk = nheads * d, kvheads * d
+        q = qkv[..., :hq].view(b, s, nheads, d)
+        k = qkv[..., hq:hq + hk].view(b, s, kvheads, d)
+        v = qkv[..., hq + hk:].view(b, s, kvheads, d)
+        cos = cos.float().contiguous()
+        sin = sin.float().contiguous()
+        qr, kr = ext.rope_fwd(q, k, cos, sin, False)   # strided reads
+        o, lse = ext.attn_fwd(qr, kr, v)               # strided v
+        ctx.save_for_backward(qkv, qr, kr, o, lse, cos, sin)
+        ctx.dims = (nheads, kvheads, d)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, qr, kr, o, lse, cos, sin = ctx.saved_tensors
+        nheads, kvheads, d = ctx.dims
+        b, s, _ = qkv.shape
+        hq, hk = nheads * d, kvheads * d
+        v = qkv[..., hq + hk:].view(b, s, kvheads, d)
+        dqkv = torch.empty_like(qkv)
+        dq_sl = dqkv[..., :hq].view(b, s, nheads, d)
+        dk_sl = dqkv[..., hq:hq + hk].view(b, s, kvheads, d)
+        dv_sl = dqkv[..., hq + hk:].view(b, s, kvheads, d)
+        if nheads == kvheads:
+            # dv written straight into the fused buffer by the kernel
+            dq, dk, _ = _C.attn_bwd(do.contiguous(), qr, kr, v, o, lse,
+                                    dv_sl)
+        else:
+            dq, dk, dv = _C.attn_bwd(do.contiguous(), qr, kr, v, o, lse,
+                                     None)
+            dv_sl.copy_(dv)
+        # inverse rotation scattered into the fused buffer slices
+        _C.rope_into(dq, dq_sl, cos, sin, True)
+        _C.rope_into(dk, dk_sl, cos, sin, True)
+        return dqkv, None, None, None, None, None
+
+
+def qkv_rope_attention(qkv, cos, sin, nheads, kvheads, head_dim):
+    """Fused GPU path; callers gate on seq%128==0 and head_dim in
+    (64, 128) and fall back to the modular split+rope+attention path
+    otherwise (decode/KV-cache, odd lengths, CPU)."""
+    return _QKVRopeAttnFn.apply(qkv, cos, sin, nheads, kvheads, head_dim)
 
 
 def attention_causal(q, k, v):

@@ -100,7 +100,8 @@ template <int D>
 __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const short* __restrict__ qg, const short* __restrict__ kg,
     const short* __restrict__ vg, short* __restrict__ og,
-    float* __restrict__ lseg, int B, int S, int H, int KVH, float scale) {
+    float* __restrict__ lseg, int B, int S, int H, int KVH, float scale,
+    long long vstride) {
   constexpr int KVB = 64;
   constexpr int NC = D / 16;   // QK^T k-chunks
   constexpr int NT = D / 32;   // 32-wide output tiles
@@ -130,7 +131,8 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   const long long krow_stride = (long long)KVH * D;
   const short* qbase = qg + ((long long)b * S * H + (long long)h) * D;
   const short* kbase = kg + ((long long)b * S * KVH + (long long)kvh) * D;
-  const short* vbase = vg + ((long long)b * S * KVH + (long long)kvh) * D;
+  // v may be a strided view (a slice of the fused qkv projection)
+  const short* vbase = vg + (long long)b * S * vstride + (long long)kvh * D;
 
   // Q -> B-fragments (registers, reused all tiles)
   bf16x8v qb[NC];
@@ -156,13 +158,14 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
     for (int c2 = 0; c2 < D / 64; ++c2) {
       const int dc = d0 + c2 * 16;
-      const long long g = (long long)(kv0 + key) * krow_stride + dc;
+      const long long gk = (long long)(kv0 + key) * krow_stride + dc;
+      const long long gv = (long long)(kv0 + key) * vstride + dc;
       const int off0 = SUBT_OFF(key, dc, 16);
       const int off1 = SUBT_OFF(key, dc + 8, 16);
-      *(f32x4*)(KLDS(buf) + off0) = *(const f32x4*)(kbase + g);
-      *(f32x4*)(KLDS(buf) + off1) = *(const f32x4*)(kbase + g + 8);
-      *(f32x4*)(VLDS(buf) + off0) = *(const f32x4*)(vbase + g);
-      *(f32x4*)(VLDS(buf) + off1) = *(const f32x4*)(vbase + g + 8);
+      *(f32x4*)(KLDS(buf) + off0) = *(const f32x4*)(kbase + gk);
+      *(f32x4*)(KLDS(buf) + off1) = *(const f32x4*)(kbase + gk + 8);
+      *(f32x4*)(VLDS(buf) + off0) = *(const f32x4*)(vbase + gv);
+      *(f32x4*)(VLDS(buf) + off1) = *(const f32x4*)(vbase + gv + 8);
     }
   };
   stage(0, 0);
@@ -307,7 +310,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const short* __restrict__ dog, const short* __restrict__ qg,
     const short* __restrict__ kg, const short* __restrict__ vg,
     const float* __restrict__ lseg, const float* __restrict__ deltag,
-    short* __restrict__ dqg, int B, int S, int H, int KVH, float scale) {
+    short* __restrict__ dqg, int B, int S, int H, int KVH, float scale,
+    long long vstride) {
   constexpr int KVB = 32;
   constexpr int NC = D / 16;
   constexpr int NT = D / 32;
@@ -337,7 +341,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   const short* qbase = qg + ((long long)b * S * H + (long long)h) * D;
   const short* dobase = dog + ((long long)b * S * H + (long long)h) * D;
   const short* kbase = kg + ((long long)b * S * KVH + (long long)kvh) * D;
-  const short* vbase = vg + ((long long)b * S * KVH + (long long)kvh) * D;
+  const short* vbase = vg + (long long)b * S * vstride + (long long)kvh * D;
 
   bf16x8v qb[NC], dob[NC];
   {
@@ -366,10 +370,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     for (int i = t256; i < 32 * CPR; i += 256) {
       const int key = i % 32;
       const int d0 = (i / 32) * 8;
-      const long long g = (long long)(kv0s + key) * krow_stride + d0;
       const int off = SUBT_OFF(key, d0, 8);
-      *(f32x4*)(KIMG(buf) + off) = *(const f32x4*)(kbase + g);
-      *(f32x4*)(VIMG(buf) + off) = *(const f32x4*)(vbase + g);
+      *(f32x4*)(KIMG(buf) + off) =
+          *(const f32x4*)(kbase + (long long)(kv0s + key) * krow_stride + d0);
+      *(f32x4*)(VIMG(buf) + off) =
+          *(const f32x4*)(vbase + (long long)(kv0s + key) * vstride + d0);
     }
   };
   stage(0, 0);
@@ -456,7 +461,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const short* __restrict__ kg, const short* __restrict__ vg,
     const float* __restrict__ lseg, const float* __restrict__ deltag,
     void* __restrict__ dkg, void* __restrict__ dvg, int B, int S, int H,
-    int KVH, float scale) {
+    int KVH, float scale, long long vstride, long long dvstride) {
   constexpr int KVB = 32;   // keys per wave; block = 4 waves = 128 keys
   constexpr int KSWZ = (D == 128) ? 15 : 7;
   constexpr int NC = D / 16;
@@ -487,7 +492,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   const short* qbase = qg + ((long long)b * S * H + (long long)h) * D;
   const short* dobase = dog + ((long long)b * S * H + (long long)h) * D;
   const short* kbase = kg + ((long long)b * S * KVH + (long long)kvh) * D;
-  const short* vbase = vg + ((long long)b * S * KVH + (long long)kvh) * D;
+  const short* vbase = vg + (long long)b * S * vstride + (long long)kvh * D;
 
   char* my_k = k_lds + wid * KVB * D * 2;
   char* my_p = p_lds + wid * KVB * 64;
@@ -507,7 +512,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   // V fragments stay in registers (A-operand rows = this lane's key)
   bf16x8v vreg[NC];
   {
-    const short* vp = vbase + (long long)(kv0 + col) * krow_stride + hb * 8;
+    const short* vp = vbase + (long long)(kv0 + col) * vstride + hb * 8;
 #pragma unroll
     for (int c = 0; c < NC; ++c) vreg[c] = *(const bf16x8v*)(vp + c * 16);
   }
@@ -644,15 +649,18 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       const long long off =
           ((long long)b * S + key_abs) * (long long)(KVH * D) +
           (long long)kvh * D + feat;
+      // dv may target a strided slice of a fused dqkv buffer
+      const long long offv = ((long long)b * S + key_abs) * dvstride +
+                             (long long)kvh * D + feat;
       if (OUT_BF16) {
         ((short*)dkg)[off] = f2bf(accDK[t][r]);
-        ((short*)dvg)[off] = f2bf(accDV[t][r]);
+        ((short*)dvg)[offv] = f2bf(accDV[t][r]);
       } else if (ngrp > 1) {
         atomicAdd((float*)dkg + off, accDK[t][r]);
-        atomicAdd((float*)dvg + off, accDV[t][r]);
+        atomicAdd((float*)dvg + offv, accDV[t][r]);
       } else {
         ((float*)dkg)[off] += accDK[t][r];
-        ((float*)dvg)[off] += accDV[t][r];
+        ((float*)dvg)[offv] += accDV[t][r];
       }
     }
   }
@@ -662,23 +670,24 @@ extern "C" {
 
 void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
                      float* lse, int B, int S, int H, int KVH, int D,
-                     float scale, hipStream_t stream) {
+                     float scale, long long vstride, hipStream_t stream) {
   dim3 grid(S / 128, B * H);
   const int lds = 2 * (64 * D * 2 + D * 64 * 2);  // dbuf K + Vt
   if (D == 128)
     attn_fwd_kernel<128><<<grid, 256, lds, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse, B,
-        S, H, KVH, scale);
+        S, H, KVH, scale, vstride);
   else
     attn_fwd_kernel<64><<<grid, 256, lds, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse, B,
-        S, H, KVH, scale);
+        S, H, KVH, scale, vstride);
 }
 
 void launch_attn_bwd(const void* do_, const void* q, const void* k,
                      const void* v, const void* o, const float* lse, void* dq,
                      void* dk, void* dv, float* delta_ws, int B, int S, int H,
                      int KVH, int D, float scale, int out_bf16,
+                     long long vstride, long long dvstride,
                      hipStream_t stream) {
   const long long rows = (long long)B * S * H;
   attn_bwd_delta_kernel<<<(int)((rows * 64 + 255) / 256), 256, 0, stream>>>(
@@ -688,30 +697,34 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
     const int lds_dq = 4 * 32 * 128 * 2;   // dbuf subtiled k+v images
     attn_bwd_dq_kernel<128><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
-        lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
+        lse, delta_ws, (short*)dq, B, S, H, KVH, scale, vstride);
     const int lds_dkv = 4 * 32 * 128 * 2 + 4 * 32 * 128 * 2 + 4 * 32 * 64;
     if (out_bf16)
       attn_bwd_dkv_kernel<128, true><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
-          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale);
+          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale,
+          vstride, dvstride);
     else
       attn_bwd_dkv_kernel<128, false><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
-          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale);
+          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale,
+          vstride, dvstride);
   } else {
     const int lds_dq = 4 * 32 * 64 * 2;
     attn_bwd_dq_kernel<64><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
-        lse, delta_ws, (short*)dq, B, S, H, KVH, scale);
+        lse, delta_ws, (short*)dq, B, S, H, KVH, scale, vstride);
     const int lds_dkv = 4 * 32 * 64 * 2 + 4 * 32 * 64 * 2 + 4 * 32 * 64;
     if (out_bf16)
       attn_bwd_dkv_kernel<64, true><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
-          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale);
+          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale,
+          vstride, dvstride);
     else
       attn_bwd_dkv_kernel<64, false><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
-          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale);
+          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale,
+          vstride, dvstride);
   }
 }
 

@@ -16,7 +16,7 @@ void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
 void launch_add_rmsnorm_fwd(const void*, const void*, const void*, void*,
                             void*, float*, int, int, float, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, int, int,
-                 int, int, int, hipStream_t);
+                 int, int, int, long long, long long, hipStream_t);
 void launch_swiglu_fwd(const void*, void*, long long, int, hipStream_t);
 void launch_swiglu_bwd(const void*, const void*, void*, long long, int,
                        hipStream_t);
@@ -27,10 +27,11 @@ void launch_adamw(float*, const void*, int, float*, float*, void*,
                   const float*, hipStream_t);
 void launch_sqnorm(const void*, int, float*, long long, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
-                     int, int, int, int, int, float, hipStream_t);
+                     int, int, int, int, int, float, long long, hipStream_t);
 void launch_attn_bwd(const void*, const void*, const void*, const void*,
                      const void*, const float*, void*, void*, void*, float*,
-                     int, int, int, int, int, float, int, hipStream_t);
+                     int, int, int, int, int, float, int, long long,
+                     long long, hipStream_t);
 void launch_cconv_fwd(const void*, const void*, const float*, void*, int,
                       int, int, int, hipStream_t);
 void launch_cconv_bwd(const void*, const void*, const void*, const float*,
@@ -93,21 +94,47 @@ std::tuple<Tensor, Tensor, Tensor> add_rmsnorm_fwd(Tensor x, Tensor res,
   return {y, s_out, rinv};
 }
 
+// (b, s, h, d) bf16 view, contiguous within a (b, s) row, uniform row
+// stride — a last-dim slice of a fused qkv projection qualifies.
+static long long row_stride4(const Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, "expected bf16");
+  TORCH_CHECK(t.dim() == 4 && t.stride(3) == 1 &&
+                  t.stride(2) == t.size(3) &&
+                  t.stride(1) >= t.size(2) * t.size(3) &&
+                  t.stride(0) == t.size(1) * t.stride(1),
+              "expected row-contiguous (b,s,h,d) view");
+  return (long long)t.stride(1);
+}
+
 std::tuple<Tensor, Tensor> rope_fwd(Tensor q, Tensor k, Tensor cos,
                                     Tensor sin, bool conj) {
-  CHECK_BF16_CONTIG(q);
-  CHECK_BF16_CONTIG(k);
   TORCH_CHECK(cos.scalar_type() == torch::kFloat32 && cos.is_contiguous());
   const int b = q.size(0), s = q.size(1), h = q.size(2), d = q.size(3);
   const int kvh = k.size(2);
+  const long long qs = row_stride4(q), ks = row_stride4(k);
   TORCH_CHECK((d / 2) % 4 == 0, "head_dim/2 must be divisible by 4");
-  auto qo = torch::empty_like(q);
-  auto ko = torch::empty_like(k);
+  auto qo = torch::empty({b, s, h, d}, q.options());
+  auto ko = torch::empty({b, s, kvh, d}, k.options());
   launch_rope(q.data_ptr(), qo.data_ptr(), cos.data_ptr<float>(),
-              sin.data_ptr<float>(), b, s, h, d, conj, cur_stream());
+              sin.data_ptr<float>(), b, s, h, d, conj, qs,
+              (long long)h * d, cur_stream());
   launch_rope(k.data_ptr(), ko.data_ptr(), cos.data_ptr<float>(),
-              sin.data_ptr<float>(), b, s, kvh, d, conj, cur_stream());
+              sin.data_ptr<float>(), b, s, kvh, d, conj, ks,
+              (long long)kvh * d, cur_stream());
   return {qo, ko};
+}
+
+// Rotate `src` (contiguous (b,s,h,d)) and scatter into `dst`, a
+// row-strided view (slice of a fused dqkv grad buffer). Used by the
+// fused qkv+rope+attention backward.
+void rope_into(Tensor src, Tensor dst, Tensor cos, Tensor sin, bool conj) {
+  const int b = src.size(0), s = src.size(1), h = src.size(2),
+            d = src.size(3);
+  const long long is = row_stride4(src), os = row_stride4(dst);
+  TORCH_CHECK(dst.size(2) == h && dst.size(3) == d);
+  launch_rope(src.data_ptr(), dst.data_ptr(), cos.data_ptr<float>(),
+              sin.data_ptr<float>(), b, s, h, d, conj, is, os,
+              cur_stream());
 }
 
 Tensor swiglu_fwd(Tensor gu) {
@@ -172,7 +199,7 @@ void sq_norm_accum(Tensor t, Tensor out) {
 std::tuple<Tensor, Tensor> attn_fwd(Tensor q, Tensor k, Tensor v) {
   CHECK_BF16_CONTIG(q);
   CHECK_BF16_CONTIG(k);
-  CHECK_BF16_CONTIG(v);
+  const long long vs = row_stride4(v);  // strided (fused-qkv slice) ok
   const int b = q.size(0), s = q.size(1), h = q.size(2), d = q.size(3);
   const int kvh = k.size(2);
   TORCH_CHECK(d == 64 || d == 128, "head_dim must be 64 or 128");
@@ -181,14 +208,16 @@ std::tuple<Tensor, Tensor> attn_fwd(Tensor q, Tensor k, Tensor v) {
   auto lse = torch::empty({b, h, s}, q.options().dtype(torch::kFloat32));
   const float scale = 1.f / sqrtf((float)d);
   launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                  lse.data_ptr<float>(), b, s, h, kvh, d, scale,
+                  lse.data_ptr<float>(), b, s, h, kvh, d, scale, vs,
                   cur_stream());
   return {o, lse};
 }
 
 std::tuple<Tensor, Tensor, Tensor> attn_bwd(Tensor do_, Tensor q, Tensor k,
-                                            Tensor v, Tensor o, Tensor lse) {
+                                            Tensor v, Tensor o, Tensor lse,
+                                            c10::optional<Tensor> dv_out) {
   CHECK_BF16_CONTIG(do_);
+  const long long vs = row_stride4(v);
   const int b = q.size(0), s = q.size(1), h = q.size(2), d = q.size(3);
   const int kvh = k.size(2);
   auto dq = torch::empty_like(q);
@@ -196,14 +225,23 @@ std::tuple<Tensor, Tensor, Tensor> attn_bwd(Tensor do_, Tensor q, Tensor k,
   auto opts = q.options().dtype(out_bf16 ? torch::kBFloat16 : torch::kFloat32);
   auto dk = out_bf16 ? torch::empty({b, s, kvh, d}, opts)
                      : torch::zeros({b, s, kvh, d}, opts);
-  auto dv = out_bf16 ? torch::empty({b, s, kvh, d}, opts)
-                     : torch::zeros({b, s, kvh, d}, opts);
+  // dv may land directly in a strided slice of a fused dqkv buffer
+  // (bf16 path only — the GQA path accumulates in fp32 scratch)
+  Tensor dv;
+  long long dvs = (long long)kvh * d;
+  if (dv_out.has_value() && out_bf16) {
+    dv = *dv_out;
+    dvs = row_stride4(dv);
+  } else {
+    dv = out_bf16 ? torch::empty({b, s, kvh, d}, opts)
+                  : torch::zeros({b, s, kvh, d}, opts);
+  }
   auto delta = torch::empty({b, h, s}, q.options().dtype(torch::kFloat32));
   const float scale = 1.f / sqrtf((float)d);
   launch_attn_bwd(do_.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                   o.data_ptr(), lse.data_ptr<float>(), dq.data_ptr(),
                   dk.data_ptr(), dv.data_ptr(), delta.data_ptr<float>(), b, s,
-                  h, kvh, d, scale, out_bf16 ? 1 : 0, cur_stream());
+                  h, kvh, d, scale, out_bf16 ? 1 : 0, vs, dvs, cur_stream());
   if (out_bf16) return {dq, dk, dv};
   return {dq, dk.to(torch::kBFloat16), dv.to(torch::kBFloat16)};
 }
@@ -270,6 +308,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("rmsnorm_bwd", &rmsnorm_bwd);
   mod.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
   mod.def("rope_fwd", &rope_fwd);
+  mod.def("rope_into", &rope_into);
   mod.def("swiglu_fwd", &swiglu_fwd);
   mod.def("swiglu_bwd", &swiglu_bwd);
   mod.def("ce_fwd_bwd", &ce_fwd_bwd);

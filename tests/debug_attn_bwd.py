@@ -28,7 +28,7 @@ def run_shape(b, s, h, kvh, d):
     v = torch.randn(b, s, kvh, d, device=dev, dtype=torch.bfloat16)
     do = torch.randn(b, s, h, d, device=dev, dtype=torch.bfloat16)
     o, lse = _C.attn_fwd(q, k, v)
-    dq, dk, dv = _C.attn_bwd(do, q, k, v, o, lse)
+    dq, dk, dv = _C.attn_bwd(do, q, k, v, o, lse, None)
     # reference per (b, h)
     rep = h // kvh
     dq_ref = torch.zeros_like(dq, dtype=torch.float32)
@@ -91,7 +91,7 @@ dq_ref = dS @ K
 dk_ref = dS.t() @ Q
 dv_ref = P.t() @ dO
 
-dq, dk, dv = _C.attn_bwd(do, q, k, v, o, lse)
+dq, dk, dv = _C.attn_bwd(do, q, k, v, o, lse, None)
 dq = dq.float()[0, :, 0]
 dk = dk.float()[0, :, 0]
 dv = dv.float()[0, :, 0]

@@ -170,7 +170,7 @@ def test_attention_bwd(b, s, h, kvh, d):
     of.backward(do.float())
 
     o, lse = _C.attn_fwd(q, k, v)
-    dq, dk, dv = _C.attn_bwd(do, q, k, v, o, lse)
+    dq, dk, dv = _C.attn_bwd(do, q, k, v, o, lse, None)
     assert relerr(dq, qf.grad) < 6e-2
     assert relerr(dk, kf.grad) < 6e-2
     assert relerr(dv, vf.grad) < 6e-2
@@ -361,6 +361,43 @@ def test_segsum_exp():
     Lf.backward(g)
     dcs = _C.segsum_exp_bwd(g.bfloat16().contiguous(), cs)
     assert relerr(dcs, csf.grad) < 5e-2
+
+
+@pytest.mark.parametrize("b,s,h,kvh,d", [
+    (1, 256, 4, 4, 128),
+    (2, 256, 8, 2, 128),
+    (1, 256, 4, 4, 64),
+])
+def test_qkv_rope_attention_fused(b, s, h, kvh, d):
+    """Fused split+RoPE+attention (strided qkv-slice kernels, fused dqkv
+    backward buffer) must match the modular split -> rope -> attention
+    composition, values and grads."""
+    from fms_fsdp_amd import ops
+    torch.manual_seed(12)
+    E = (h + 2 * kvh) * d
+    qkv = torch.randn(b, s, E, device=dev(), dtype=torch.bfloat16)
+    pos = torch.arange(s, dtype=torch.float32)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, d, 2,
+                                          dtype=torch.float32) / d))
+    ang = torch.outer(pos, inv)
+    cos, sin = ang.cos().to(dev()), ang.sin().to(dev())
+
+    a = qkv.clone().requires_grad_()
+    o_f = ops.qkv_rope_attention(a, cos, sin, h, kvh, d)
+    do = torch.randn_like(o_f)
+    o_f.backward(do)
+
+    b2 = qkv.clone().requires_grad_()
+    q, k, v = b2.split([h * d, kvh * d, kvh * d], dim=-1)
+    q = q.view(b, s, h, d)
+    k = k.view(b, s, kvh, d)
+    v = v.view(b, s, kvh, d)
+    qr, kr = ops.rope_apply(q, k, cos, sin)
+    o_m = ops.attention_causal(qr, kr, v)
+    o_m.backward(do)
+
+    assert relerr(o_f, o_m) < 1e-2
+    assert relerr(a.grad, b2.grad) < 2e-2
 
 
 def test_attention_padded_seq():

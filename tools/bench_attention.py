@@ -39,7 +39,7 @@ def main():
               f"{fwd_flops/t/1e12:7.1f} TF/s")
 
         bwd_flops = fwd_flops * 2.5
-        t = bench(lambda: _C.attn_bwd(do, q, k, v, o, lse), iters=10)
+        t = bench(lambda: _C.attn_bwd(do, q, k, v, o, lse, None), iters=10)
         print(f"bwd  b{b} s{s} h{h} kvh{kvh}: {t*1e3:8.3f} ms  "
               f"{bwd_flops/t/1e12:7.1f} TF/s (dq+dkv+delta)")
 
