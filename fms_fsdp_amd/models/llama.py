@@ -120,10 +120,26 @@ class Attention(nn.Module):
         qkv_out = (cfg.nheads + 2 * cfg.kvheads) * cfg.head_dim
         self.qkv = nn.Linear(cfg.emb_dim, qkv_out, bias=False)
         self.proj = nn.Linear(cfg.emb_dim, cfg.emb_dim, bias=False)
+        # wgrads accumulate straight into the sharded runtime's flat grad
+        # buffer when wrapped (ops.linear_flat; no-op for plain models)
+        self.qkv.weight._direct_wgrad = True
+        self.proj.weight._direct_wgrad = True
+
+    def _project_out(self, o2, residual, b, s):
+        """Output projection; residual add fused into the GEMM epilogue.
+        TP (_disable_fused_residual) must go through the proj MODULE —
+        its all-reduce hook is registered on it."""
+        if getattr(self, "_disable_fused_residual", False):
+            out = self.proj(o2.reshape(b * s, -1)).view(b, s, -1)
+            return out if residual is None else out + residual
+        return ops.linear_flat(o2.view(b, s, -1), self.proj.weight, residual)
 
     def forward(self, x, cos, sin, cache=None, residual=None):
         b, s, _ = x.shape
-        qkv = self.qkv(x)
+        if getattr(self, "_disable_fused_residual", False):
+            qkv = self.qkv(x)
+        else:
+            qkv = ops.linear_flat(x, self.qkv.weight)
         if (cache is None and qkv.is_cuda and s % 128 == 0
                 and self.head_dim in (64, 128)
                 and qkv.dtype == torch.bfloat16
@@ -134,13 +150,7 @@ class Attention(nn.Module):
             # fused dqkv buffer (no cat/copies)
             o = ops.qkv_rope_attention(qkv, cos, sin, self.nheads,
                                        self.kvheads, self.head_dim)
-            o2 = o.reshape(b * s, -1)
-            if residual is not None and \
-                    not getattr(self, "_disable_fused_residual", False):
-                return torch.addmm(residual.reshape(b * s, -1), o2,
-                                   self.proj.weight.t()).view(b, s, -1)
-            out = self.proj(o2).view(b, s, -1)
-            return out if residual is None else out + residual
+            return self._project_out(o.reshape(b, s, -1), residual, b, s)
         q, k, v = qkv.split(
             [self.nheads * self.head_dim,
              self.kvheads * self.head_dim,
@@ -163,14 +173,7 @@ class Attention(nn.Module):
                 enable_gqa=(self.kvheads != self.nheads)).transpose(1, 2)
         else:
             o = ops.attention_causal(q, k, v)      # (b, s, nheads, head_dim)
-        o2 = o.reshape(b * s, -1)
-        if residual is not None and \
-                not getattr(self, "_disable_fused_residual", False):
-            # residual add fused into the output-projection GEMM epilogue
-            return torch.addmm(residual.reshape(b * s, -1), o2,
-                               self.proj.weight.t()).view(b, s, -1)
-        out = self.proj(o2).view(b, s, -1)
-        return out if residual is None else out + residual
+        return self._project_out(o.reshape(b, s, -1), residual, b, s)
 
     def reset_parameters(self):
         for lin in (self.qkv, self.proj):
@@ -183,18 +186,18 @@ class SwiGLU(nn.Module):
         self.hidden_dim = cfg.hidden_dim
         self.wg1 = nn.Linear(cfg.emb_dim, 2 * cfg.hidden_dim, bias=False)  # fused gate|up
         self.w2 = nn.Linear(cfg.hidden_dim, cfg.emb_dim, bias=False)
+        self.wg1.weight._direct_wgrad = True
+        self.w2.weight._direct_wgrad = True
 
     def forward(self, x, residual=None):
-        gu = self.wg1(x)
+        if getattr(self, "_disable_fused_residual", False):   # TP path
+            gu = self.wg1(x)
+            h = ops.swiglu(gu)
+            out = self.w2(h)
+            return out if residual is None else out + residual.view_as(out)
+        gu = ops.linear_flat(x, self.wg1.weight)
         h = ops.swiglu(gu)                         # silu(g) * u, fused kernel
-        if residual is not None and \
-                not getattr(self, "_disable_fused_residual", False):
-            b, s, e = residual.shape
-            return torch.addmm(residual.reshape(b * s, e),
-                               h.reshape(b * s, -1),
-                               self.w2.weight.t()).view(b, s, e)
-        out = self.w2(h)
-        return out if residual is None else out + residual.view_as(out)
+        return ops.linear_flat(h, self.w2.weight, residual)
 
     def reset_parameters(self):
         for lin in (self.wg1, self.w2):

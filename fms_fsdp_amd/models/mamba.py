@@ -163,6 +163,9 @@ class Mamba2Mixer(nn.Module):
         self.D = nn.Parameter(torch.ones(self.nheads))
         self.norm = RMSNorm(self.d_inner, eps=1e-5)
         self.out_proj = nn.Linear(self.d_inner, self.d_model, bias=False)
+        # wgrads land directly in the sharded runtime's flat grad buffer
+        self.in_proj.weight._direct_wgrad = True
+        self.out_proj.weight._direct_wgrad = True
 
     def reset_parameters(self):
         nn.init.trunc_normal_(self.in_proj.weight, std=0.02)
@@ -184,7 +187,7 @@ class Mamba2Mixer(nn.Module):
 
     def forward(self, u):
         b, l, _ = u.shape
-        zxbcdt = self.in_proj(u)
+        zxbcdt = ops.linear_flat(u, self.in_proj.weight)
         z, xBC, dt = torch.split(
             zxbcdt, [self.d_inner, self.conv_dim, self.nheads], dim=-1)
 
@@ -216,7 +219,7 @@ class Mamba2Mixer(nn.Module):
             y = _scan(x, dt, self.dt_bias, self.A_log, B, C, self.D)
         y = y.reshape(b, l, self.d_inner).to(u.dtype)
         y = self.norm(y * F.silu(z))
-        return self.out_proj(y)
+        return ops.linear_flat(y, self.out_proj.weight)
 
 
 class MambaAttnMixer(nn.Module):
@@ -237,6 +240,10 @@ class MambaAttnMixer(nn.Module):
                               bias=ac.get("out_proj_bias", False))
         if self.rot_dim:
             self.rot_emb = RotaryEmbedding(self.rot_dim, 4096, 10000.0)
+        if self.qkv.bias is None:
+            self.qkv.weight._direct_wgrad = True
+        if self.proj.bias is None:
+            self.proj.weight._direct_wgrad = True
 
     def reset_parameters(self):
         nn.init.trunc_normal_(self.qkv.weight, std=0.02)
@@ -248,7 +255,9 @@ class MambaAttnMixer(nn.Module):
 
     def forward(self, x):
         b, s, _ = x.shape
-        q, k, v = self.qkv(x).split(
+        qkv = (self.qkv(x) if self.qkv.bias is not None
+               else ops.linear_flat(x, self.qkv.weight))
+        q, k, v = qkv.split(
             [self.nheads * self.head_dim, self.kvheads * self.head_dim,
              self.kvheads * self.head_dim], dim=-1)
         q = q.view(b, s, self.nheads, self.head_dim)
@@ -262,7 +271,10 @@ class MambaAttnMixer(nn.Module):
             q = torch.cat([qr, q[..., self.rot_dim:]], dim=-1)
             k = torch.cat([kr, k[..., self.rot_dim:]], dim=-1)
         o = ops.attention_causal(q.contiguous(), k.contiguous(), v.contiguous())
-        return self.proj(o.reshape(b, s, -1))
+        o2 = o.reshape(b, s, -1)
+        if self.proj.bias is not None:
+            return self.proj(o2)
+        return ops.linear_flat(o2, self.proj.weight)
 
 
 class GatedMLP(nn.Module):
@@ -272,13 +284,16 @@ class GatedMLP(nn.Module):
         super().__init__()
         self.wg1 = nn.Linear(d_model, 2 * d_intermediate, bias=False)
         self.w2 = nn.Linear(d_intermediate, d_model, bias=False)
+        self.wg1.weight._direct_wgrad = True
+        self.w2.weight._direct_wgrad = True
 
     def reset_parameters(self):
         nn.init.trunc_normal_(self.wg1.weight, std=0.02)
         nn.init.trunc_normal_(self.w2.weight, std=0.02)
 
     def forward(self, x):
-        return self.w2(ops.swiglu(self.wg1(x)))
+        gu = ops.linear_flat(x, self.wg1.weight)
+        return ops.linear_flat(ops.swiglu(gu), self.w2.weight)
 
 
 class MambaBlock(nn.Module):

@@ -131,7 +131,15 @@ class FlatUnit:
 
     def _set_grad_views(self):
         for p, off in zip(self.params, self.offsets):
-            p.grad = self.flat_grad[off:off + p.numel()].view(p.shape)
+            view = self.flat_grad[off:off + p.numel()].view(p.shape)
+            if getattr(p, "_direct_wgrad", False):
+                # model routes this weight through ops.linear_flat: the
+                # wgrad GEMM accumulates straight into the flat buffer
+                # (no autograd temp + add pass); p.grad stays None
+                p._flat_grad_view = view
+                p.grad = None
+            else:
+                p.grad = view
         self._grads_ready_views = True
 
     # ---------------- storage control ----------------
@@ -323,7 +331,14 @@ class ShardedModel(nn.Module):
             u.module.register_full_backward_pre_hook(self._make_bwd_pre(i))
         for u in self.all_units:
             for p in u.params:
-                p.register_post_accumulate_grad_hook(self._make_grad_hook(u))
+                if getattr(p, "_direct_wgrad", False):
+                    # ops.linear_flat's backward fires this after its
+                    # addmm_ into the flat buffer — same countdown as
+                    # the post-accumulate hook of ordinary params
+                    p._wgrad_done = self._make_wgrad_cb(u)
+                else:
+                    p.register_post_accumulate_grad_hook(
+                        self._make_grad_hook(u))
         # no separate root unit: nothing extra to gather at forward start
 
     def _make_fwd_pre(self, idx):
@@ -359,16 +374,24 @@ class ShardedModel(nn.Module):
             return None
         return hook
 
+    def _dec_grad(self, u: FlatUnit):
+        if u._grad_countdown == 0:
+            u._grad_countdown = len(u.params)
+        u._grad_countdown -= 1
+        if u._grad_countdown == 0:
+            u.reduce_grads(self.rs_stream, self.reduce_dtype)
+            if u.reshard_after_forward:
+                u._free_flat_param()
+
     def _make_grad_hook(self, u: FlatUnit):
         def hook(param):
-            if u._grad_countdown == 0:
-                u._grad_countdown = len(u.params)
-            u._grad_countdown -= 1
-            if u._grad_countdown == 0:
-                u.reduce_grads(self.rs_stream, self.reduce_dtype)
-                if u.reshard_after_forward:
-                    u._free_flat_param()
+            self._dec_grad(u)
         return hook
+
+    def _make_wgrad_cb(self, u: FlatUnit):
+        def cb():
+            self._dec_grad(u)
+        return cb
 
     def forward(self, *args, **kwargs):
         # root params (embedding) needed first; gather root + lookahead
