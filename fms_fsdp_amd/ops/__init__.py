@@ -148,6 +148,7 @@ class _DirectWgradLinearFn(torch.autograd.Function):
         # attribute holders stashed at forward time: saved_tensors can
         # return attr-less re-wrapped tensors (e.g. under AC recompute)
         ctx.gview = weight._flat_grad_view
+        ctx.wobj = weight
         ctx.cb = getattr(weight, "_wgrad_done", None)
         ctx.res_shape = residual.shape if residual is not None else None
         x2 = x.reshape(-1, x.shape[-1])
@@ -164,9 +165,14 @@ class _DirectWgradLinearFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
         dx = (dy2 @ w).view(x.shape)
-        # wgrad straight into the flat grad buffer (+= for accumulation;
-        # the buffer is zeroed once per optimizer step)
-        ctx.gview.addmm_(dy2.t(), x2)
+        # wgrad straight into the flat grad buffer. First touch since
+        # zero_grad overwrites (beta=0) so the runtime never has to
+        # zero-fill these slices; later touches accumulate (beta=1).
+        if getattr(ctx.wobj, "_wgrad_fresh", False):
+            torch.addmm(ctx.gview, dy2.t(), x2, beta=0, out=ctx.gview)
+            ctx.wobj._wgrad_fresh = False
+        else:
+            ctx.gview.addmm_(dy2.t(), x2)
         if ctx.cb is not None:
             ctx.cb()
         dres = dy.reshape(ctx.res_shape) if ctx.res_shape is not None else None

@@ -421,7 +421,21 @@ class ShardedModel(nn.Module):
     def zero_grad(self, set_to_none=False):
         self._clip_coef = None
         for u in self.all_units:
-            u.flat_grad.zero_()
+            # direct-wgrad slices are never pre-zeroed: their first
+            # wgrad of the step overwrites (beta=0 addmm). Only the
+            # ordinary-param slices (norm weights, embeddings) need the
+            # zero-fill for autograd accumulation. Alignment padding was
+            # zeroed at materialize() and is never written afterwards.
+            direct = [p for p in u.params
+                      if getattr(p, "_direct_wgrad", False)]
+            if not direct:
+                u.flat_grad.zero_()
+            else:
+                for p, off in zip(u.params, u.offsets):
+                    if getattr(p, "_direct_wgrad", False):
+                        p._wgrad_fresh = True
+                    else:
+                        u.flat_grad[off:off + p.numel()].zero_()
             if u.S > 1:
                 u.grad_shard.zero_()
             if not u._grads_ready_views:
