@@ -91,6 +91,34 @@ __device__ __forceinline__ bf16x8v tr_read2(const char* base, int off0,
   return r.v;
 }
 
+// Four transpose reads -> TWO A-fragments behind ONE lgkmcnt wait: the
+// LDS pipe streams all four b64 reads before the single serialization
+// point, so back-to-back mfmas consume both fragments densely.
+struct TR4 {
+  bf16x8v a, b;
+};
+__device__ __forceinline__ TR4 tr_read4(const char* base, int off0, int off1,
+                                        int off2, int off3) {
+  U2x64 r0, r1;
+  const unsigned a0 = (unsigned)(unsigned long long)(base + off0);
+  const unsigned a1 = (unsigned)(unsigned long long)(base + off1);
+  const unsigned a2 = (unsigned)(unsigned long long)(base + off2);
+  const unsigned a3 = (unsigned)(unsigned long long)(base + off3);
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n\t"
+      "ds_read_b64_tr_b16 %1, %5\n\t"
+      "ds_read_b64_tr_b16 %2, %6\n\t"
+      "ds_read_b64_tr_b16 %3, %7\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(r0.u[0]), "=&v"(r0.u[1]), "=&v"(r1.u[0]), "=&v"(r1.u[1])
+      : "v"(a0), "v"(a1), "v"(a2), "v"(a3)
+      : "memory");
+  TR4 out;
+  out.a = r0.v;
+  out.b = r1.v;
+  return out;
+}
+
 
 // ---------------------------------------------------------------------
 // Forward. Block = 4 waves x 32 q-rows = 128 q rows; KV tile = 64 keys.
@@ -229,19 +257,24 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
         const int dvblk = t * 2 + (col >> 4);
-#pragma unroll
-        for (int kc2 = 0; kc2 < 2; ++kc2) {
-          const int key0 = kt * 32 + kc2 * 16 + hb * 8;
-          const int rg0 = key0 >> 2, rg1 = (key0 + 4) >> 2;
-          const bf16x8v a = tr_read2(
-              VLDS(cur),
-              (dvblk * (KVB / 4) + rg0) * 128 +
-                  (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
-              (dvblk * (KVB / 4) + rg1) * 128 +
-                  (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
-          accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              a, kc2 ? pb1 : pb0, accO[t], 0, 0, 0);
-        }
+        const int ka = kt * 32 + hb * 8;        // kc2 = 0 chunk
+        const int kb = kt * 32 + 16 + hb * 8;   // kc2 = 1 chunk
+        const int rga0 = ka >> 2, rga1 = (ka + 4) >> 2;
+        const int rgb0 = kb >> 2, rgb1 = (kb + 4) >> 2;
+        const TR4 f = tr_read4(
+            VLDS(cur),
+            (dvblk * (KVB / 4) + rga0) * 128 +
+                (((lane & 15) * 8) ^ ((rga0 & 2) << 3)),
+            (dvblk * (KVB / 4) + rga1) * 128 +
+                (((lane & 15) * 8) ^ ((rga1 & 2) << 3)),
+            (dvblk * (KVB / 4) + rgb0) * 128 +
+                (((lane & 15) * 8) ^ ((rgb0 & 2) << 3)),
+            (dvblk * (KVB / 4) + rgb1) * 128 +
+                (((lane & 15) * 8) ^ ((rgb1 & 2) << 3)));
+        accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(f.a, pb0, accO[t],
+                                                          0, 0, 0);
+        accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(f.b, pb1, accO[t],
+                                                          0, 0, 0);
       }
     }
     __syncthreads();
@@ -421,18 +454,19 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 #pragma unroll
     for (int t = 0; t < NT; ++t) {
       const int dkblk = t * 2 + (col >> 4);
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        const int key0 = kc * 16 + hb * 8;
-        const int rg0 = key0 >> 2, rg1 = (key0 + 4) >> 2;
-        const bf16x8v a = tr_read2(
-            KIMG(cur),
-            (dkblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
-            (dkblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
-        accDQ[t] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, dsb[kc], accDQ[t],
-                                                    0, 0, 0);
-      }
+      const int ka = hb * 8, kb = 16 + hb * 8;
+      const int rga0 = ka >> 2, rga1 = (ka + 4) >> 2;
+      const int rgb0 = kb >> 2, rgb1 = (kb + 4) >> 2;
+      const TR4 f = tr_read4(
+          KIMG(cur),
+          (dkblk * 8 + rga0) * 128 + (((lane & 15) * 8) ^ ((rga0 & 2) << 3)),
+          (dkblk * 8 + rga1) * 128 + (((lane & 15) * 8) ^ ((rga1 & 2) << 3)),
+          (dkblk * 8 + rgb0) * 128 + (((lane & 15) * 8) ^ ((rgb0 & 2) << 3)),
+          (dkblk * 8 + rgb1) * 128 + (((lane & 15) * 8) ^ ((rgb1 & 2) << 3)));
+      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(f.a, dsb[0],
+                                                         accDQ[t], 0, 0, 0);
+      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(f.b, dsb[1],
+                                                         accDQ[t], 0, 0, 0);
     }
     __syncthreads();
     cur ^= 1;
@@ -606,17 +640,19 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
         const int dvblk = t * 2 + (col >> 4);
-#pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
-          const int qg0 = kc * 16 + hb * 8;
-          const int rg0 = qg0 >> 2, rg1 = (qg0 + 4) >> 2;
-          const bf16x8v bb = tr_read2(
-              DOIMG(cur),
-              (dvblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
-              (dvblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
-          accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kc], bb,
-                                                             accDV[t], 0, 0, 0);
-        }
+        const int qa = hb * 8, qb2 = 16 + hb * 8;
+        const int rga0 = qa >> 2, rga1 = (qa + 4) >> 2;
+        const int rgb0 = qb2 >> 2, rgb1 = (qb2 + 4) >> 2;
+        const TR4 f = tr_read4(
+            DOIMG(cur),
+            (dvblk * 8 + rga0) * 128 + (((lane & 15) * 8) ^ ((rga0 & 2) << 3)),
+            (dvblk * 8 + rga1) * 128 + (((lane & 15) * 8) ^ ((rga1 & 2) << 3)),
+            (dvblk * 8 + rgb0) * 128 + (((lane & 15) * 8) ^ ((rgb0 & 2) << 3)),
+            (dvblk * 8 + rgb1) * 128 + (((lane & 15) * 8) ^ ((rgb1 & 2) << 3)));
+        accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], f.a,
+                                                           accDV[t], 0, 0, 0);
+        accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], f.b,
+                                                           accDV[t], 0, 0, 0);
       }
     }
     // dS -> my_p, then dK[key][dk] += dS(A) @ tr(imgq)(B)
@@ -635,17 +671,19 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
         const int dkblk = t * 2 + (col >> 4);
-#pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
-          const int qg0 = kc * 16 + hb * 8;
-          const int rg0 = qg0 >> 2, rg1 = (qg0 + 4) >> 2;
-          const bf16x8v bb = tr_read2(
-              QIMG(cur),
-              (dkblk * 8 + rg0) * 128 + (((lane & 15) * 8) ^ ((rg0 & 2) << 3)),
-              (dkblk * 8 + rg1) * 128 + (((lane & 15) * 8) ^ ((rg1 & 2) << 3)));
-          accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[kc], bb,
-                                                             accDK[t], 0, 0, 0);
-        }
+        const int qa = hb * 8, qb2 = 16 + hb * 8;
+        const int rga0 = qa >> 2, rga1 = (qa + 4) >> 2;
+        const int rgb0 = qb2 >> 2, rgb1 = (qb2 + 4) >> 2;
+        const TR4 f = tr_read4(
+            QIMG(cur),
+            (dkblk * 8 + rga0) * 128 + (((lane & 15) * 8) ^ ((rga0 & 2) << 3)),
+            (dkblk * 8 + rga1) * 128 + (((lane & 15) * 8) ^ ((rga1 & 2) << 3)),
+            (dkblk * 8 + rgb0) * 128 + (((lane & 15) * 8) ^ ((rgb0 & 2) << 3)),
+            (dkblk * 8 + rgb1) * 128 + (((lane & 15) * 8) ^ ((rgb1 & 2) << 3)));
+        accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[0], f.a,
+                                                           accDK[t], 0, 0, 0);
+        accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[1], f.b,
+                                                           accDK[t], 0, 0, 0);
       }
     }
     __syncthreads();
