@@ -175,6 +175,9 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
   for (int t = 0; t < NT; ++t) accO[t] = (f32x16)(0.f);
   float m_run = -1e30f, l_run = 0.f;
+  // softmax runs in base 2 (v_exp_f32 is natively 2^x; folding log2(e)
+  // into the scale drops one v_mul per exponential)
+  const float scale2 = scale * 1.4426950408889634f;
 
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
   const int t256 = threadIdx.x;
@@ -206,7 +209,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     // ---- two 32-key sub-tiles, each with its own online-softmax pass.
     // Register economy: one live accS/p set (16 regs) instead of two,
     // keeping total VGPR+AGPR under 256 for 2 waves/SIMD occupancy.
-    // defer-max (guide T13, THR=8): the O/l rescale runs only when the
+    // defer-max (guide T13, THR=8 nats = 11.5 bits): the O/l rescale runs only when the
     // sub-tile max exceeds the running max by more than THR; P is then
     // bounded by e^THR which the fp32 accumulate tolerates. Decision is
     // made BEFORE this sub-tile's P is exponentiated (the safe order).
@@ -227,15 +230,15 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       float mt = -1e30f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        float sv = accS[r] * scale;
+        float sv = accS[r] * scale2;
         if (partial && (kv32 + DROW(r, hb)) > my_q) sv = -1e30f;
         p[r] = sv;
         mt = fmaxf(mt, sv);
       }
       mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
       float alpha = 1.f;
-      if (mt > m_run + 8.f) {          // defer-max threshold
-        alpha = __expf(m_run - mt);
+      if (mt > m_run + 11.5f) {        // defer-max threshold (base-2)
+        alpha = exp2f(m_run - mt);
         m_run = mt;
 #pragma unroll
         for (int t = 0; t < NT; ++t)
@@ -245,7 +248,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       float s_own = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        p[r] = __expf(p[r] - m_run);
+        p[r] = exp2f(p[r] - m_run);
         s_own += p[r];
       }
       l_run = l_run * alpha + s_own + __shfl_xor(s_own, 32, 64);
@@ -297,8 +300,9 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       *(bf16x4*)(op + dv0) = w;
     }
   }
-  if (hb == 0)
-    lseg[((long long)bh) * S + my_q] = m_run + __logf(l_run);
+  if (hb == 0)   // exported in natural-log units (tests, bwd contract)
+    lseg[((long long)bh) * S + my_q] =
+        (m_run + __log2f(l_run)) * 0.6931471805599453f;
 }
 #undef KLDS
 #undef VLDS
@@ -386,8 +390,10 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       dob[c] = *(const bf16x8v*)(dp + c * 16);
     }
   }
-  const float my_lse = lseg[((long long)bh) * S + my_q];
+  const float my_lse2 =
+      lseg[((long long)bh) * S + my_q] * 1.4426950408889634f;
   const float my_delta = deltag[((long long)bh) * S + my_q];
+  const float scale2 = scale * 1.4426950408889634f;
 
   f32x16 accDQ[NT];
 #pragma unroll
@@ -441,7 +447,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key = kv0 + DROW(r, hb);
-      const float pval = __expf(accS[r] * scale - my_lse);
+      const float pval = exp2f(accS[r] * scale2 - my_lse2);
       float v = pval * (accDP[r] - my_delta) * scale;
       if (key > my_q) v = 0.f;
       ds[r] = v;
@@ -611,14 +617,16 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
                                                       0, 0, 0);
     }
 
-    const float lse_q = lseg[((long long)bh) * S + q0 + col];
+    const float lse_q2 =
+        lseg[((long long)bh) * S + q0 + col] * 1.4426950408889634f;
     const float delta_q = deltag[((long long)bh) * S + q0 + col];
+    const float scale2 = scale * 1.4426950408889634f;
     float pv[16], ds[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key = kv0 + DROW(r, hb);
       const int q = q0 + col;
-      float pval = __expf(accS[r] * scale - lse_q);
+      float pval = exp2f(accS[r] * scale2 - lse_q2);
       if (key > q) pval = 0.f;
       pv[r] = pval;
       ds[r] = pval * (accDP[r] - delta_q) * scale;
