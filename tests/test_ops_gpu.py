@@ -203,6 +203,37 @@ def test_model_gpu_step():
     assert all(l == l for l in losses), losses
 
 
+def test_model_gpu_step_selective_ac():
+    """Selective AC recompute + direct-wgrad (linear_flat) + the fused
+    qkv path together on GPU: AC re-runs the block forward during
+    backward; each weight's wgrad must still fire exactly once."""
+    from fms_fsdp_amd.models import Llama, LlamaBlock, LlamaConfig
+    from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+    from fms_fsdp_amd.parallel.policies import apply_selective_ac
+    torch.manual_seed(0)
+    cfg = LlamaConfig(src_vocab_size=512, emb_dim=256, nheads=2, kvheads=2,
+                      nlayers=4, max_expected_seq_len=256)
+    with torch.device(dev()):
+        m = Llama(cfg)
+        m.reset_parameters()
+    apply_selective_ac(m, LlamaBlock, 0.5)
+    sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                      param_dtype=torch.bfloat16)
+    opt = ShardedAdamW(sm, lr=1e-3)
+    x = torch.randint(0, 512, (2, 256), device=dev())
+    y = torch.randint(0, 512, (2, 256), device=dev())
+    losses = []
+    for _ in range(5):
+        opt.zero_grad()
+        loss = sm(x, labels=y)
+        loss.backward()
+        sm.clip_grad_norm_(1.0)
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+    assert all(l == l for l in losses), losses
+
+
 def test_causal_conv1d():
     torch.manual_seed(7)
     from fms_fsdp_amd import _C
