@@ -59,6 +59,9 @@ class train_config:
 
     # logging
     report_interval: int = 100
+    # write checkpoint files from a background thread (D2H snapshot is
+    # still synchronous; metadata.pth written last = validity marker)
+    async_checkpoint: bool = False
     checkpoint_interval: int = 10000
     tracker: Optional[str] = None             # None | "wandb" | "aim"
     tracker_dir: str = "/fsx/aim_logs/llama"

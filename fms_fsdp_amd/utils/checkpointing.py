@@ -22,6 +22,7 @@ buffer from the old shard set and re-slices for the new S.
 
 import os
 import shutil
+import threading
 import time
 
 import torch
@@ -71,7 +72,7 @@ class Checkpointer:
     and metadata (reference Checkpointer: checkpointing_utils.py:65-316)."""
 
     def __init__(self, ckpt_dir, n_to_save, parallel_mode, rank, local_rank,
-                 report_fn=None):
+                 report_fn=None, async_save=False):
         self.max_ckps = n_to_save
         self.rank = rank
         self.local_rank = local_rank
@@ -79,6 +80,20 @@ class Checkpointer:
         assert parallel_mode in ("fsdp", "hsdp", "ddp")
         self.parallel_mode = parallel_mode
         self.report = report_fn if report_fn is not None else self._default_report
+        # async_save: the D2H copies happen synchronously inside save()
+        # (they must precede the next optimizer step mutating the
+        # shards), the file serialization runs in a background thread so
+        # training resumes immediately. metadata.pth — the validity
+        # marker load() checks — is written LAST, so a crash mid-write
+        # leaves an ignorable partial checkpoint, same as sync mode.
+        self.async_save = async_save
+        self._pending = None
+
+    def wait(self):
+        """Join an in-flight async checkpoint write (no-op otherwise)."""
+        if self._pending is not None:
+            self._pending.join()
+            self._pending = None
 
     def _default_report(self, output_path=None, **kwargs):
         if self.rank == 0:
@@ -136,32 +151,50 @@ class Checkpointer:
     # ------------- save -------------
 
     def save(self, step, model, optimizer, dataloader, tokens_seen=0):
+        self.wait()   # one async write in flight at a time
         t0 = time.time()
         shard_rank, S, is_writer = self._shard_info(model)
         out = os.path.join(self.ckpt_dir, f"step_{step}_ckp")
         os.makedirs(out, exist_ok=True)
         if dist.is_initialized():
             dist.barrier()
+        # D2H snapshots first (before training mutates the shards), then
+        # serialize — in a background thread when async_save is on.
+        payloads = []
         if is_writer:
             model_sd = {u.name: u.master_shard.cpu() for u in model.all_units}
-            torch.save(model_sd, os.path.join(out, f"model_{shard_rank}_of_{S}.pth"))
+            payloads.append((model_sd,
+                             os.path.join(out, f"model_{shard_rank}_of_{S}.pth")))
             opt_sd = {"step": optimizer.step_count, "lr": optimizer.param_groups[0]["lr"],
                       "units": {u.name: {"exp_avg": u.exp_avg.cpu(),
                                          "exp_avg_sq": u.exp_avg_sq.cpu()}
                                 for u in model.all_units}}
-            torch.save(opt_sd, os.path.join(out, f"optim_{shard_rank}_of_{S}.pth"))
+            payloads.append((opt_sd,
+                             os.path.join(out, f"optim_{shard_rank}_of_{S}.pth")))
         if dataloader is not None and hasattr(dataloader.dataset, "state_dict"):
-            torch.save(dataloader.dataset.state_dict(),
-                       os.path.join(out, f"loader_state_{self.rank}.pth"))
+            payloads.append((dataloader.dataset.state_dict(),
+                             os.path.join(out, f"loader_state_{self.rank}.pth")))
         if self.rank == 0:
-            torch.save({"step": step, "tokens_seen": tokens_seen,
-                        "shard_world": S, "layout": self._layout(model)},
-                       os.path.join(out, "metadata.pth"))
-        if dist.is_initialized():
-            dist.barrier()
+            # metadata last: it is the validity marker load() looks for
+            payloads.append(({"step": step, "tokens_seen": tokens_seen,
+                              "shard_world": S, "layout": self._layout(model)},
+                             os.path.join(out, "metadata.pth")))
+        if self.async_save:
+            self._pending = threading.Thread(
+                target=self._write_payloads, args=(payloads, out, t0),
+                daemon=True)
+            self._pending.start()
+        else:
+            self._write_payloads(payloads, out, t0)
+            if dist.is_initialized():
+                dist.barrier()
+        return out
+
+    def _write_payloads(self, payloads, out, t0):
+        for obj, path in payloads:
+            torch.save(obj, path)
         self.report(output_path=out, time_taken=f"{time.time() - t0:.2f}s")
         self._cleanup()
-        return out
 
     # ------------- load -------------
 
@@ -170,6 +203,7 @@ class Checkpointer:
         """Auto-discovering load. Prefers a checkpoint in the save dir (job
         resume) over `path` (continued pretraining, step reset)
         (reference: checkpointing_utils.py:184-281)."""
+        self.wait()
         save_dir_ckpt = self._validate_ckp_path(self.ckpt_dir)
         if save_dir_ckpt is not None:
             load_path = save_dir_ckpt

@@ -186,6 +186,8 @@ def train(cfg, model, local_rank, rank, train_loader, optimizer, scheduler,
             checkpointer.save(batch_idx, model, optimizer, train_loader,
                               tokens_seen=tokens_seen)
 
+    if checkpointer is not None:
+        checkpointer.wait()   # join any in-flight async checkpoint write
     if rank == 0:
         total = time.time() - start
         print(f"training done: {total:.1f}s")

@@ -117,7 +117,8 @@ def main(**kwargs):
                              weight_decay=0.1)
 
     checkpointer = Checkpointer(cfg.ckpt_save_path, 1000, cfg.sharding_strategy,
-                                rank, local_rank)
+                                rank, local_rank,
+                                async_save=cfg.async_checkpoint)
     model, optimizer, _, start_step, tokens_seen, is_resuming = checkpointer.load(
         model, optimizer,
         None if cfg.use_dummy_dataset else train_loader,
