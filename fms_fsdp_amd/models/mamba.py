@@ -95,32 +95,41 @@ def ssd_chunked(x, dt, A, B, C, chunk):
     # the vectorized path (measured ~25x slower)
     dA_cs = dA.cumsum(-1)                            # (b,nc,h,Q)
 
-    # expand B/C over head groups
-    Bh = Bc.repeat_interleave(rep, dim=3) if g != h else Bc    # (b,nc,Q,h,n)
-    Ch = Cc.repeat_interleave(rep, dim=3) if g != h else Cc
-
     # matmul-shaped work runs in bf16 with fp32 accumulation (MFMA rate;
     # mamba_ssm's own kernels take bf16 x/B/C the same way) — decay and
     # cumsum terms stay fp32. On CPU keep fp32 (bf16 matmul is slow there).
     mm_dtype = torch.bfloat16 if x.is_cuda else torch.float32
+    Bm = Bc.to(mm_dtype)
+    Cm = Cc.to(mm_dtype)
 
     # diagonal block: Y[i] = sum_{j<=i} C_i.B_j exp(dA[i]-dA[j]) dt_j x_j
-    # the (Q x Q)-sized intermediates dominate HBM traffic: L is emitted
-    # bf16 in ONE fused kernel pass (ops/hip/ssd.hip) and the scores /
-    # masked product stay in mm_dtype end-to-end
+    # scores C.B^T are PER GROUP (g << h; g = 1 for the registry
+    # configs): computing them per query-head — or materializing
+    # repeat_interleave'd (b,nc,Q,h,n) copies of B/C — wastes (h/g)x
+    # flops and HBM. The (Q x Q)-sized L is emitted bf16 in ONE fused
+    # kernel pass (ops/hip/ssd.hip).
     from fms_fsdp_amd.ops import segsum_exp
     L = segsum_exp(dA_cs).to(mm_dtype)               # (b,nc,h,Q,Q)
-    scores = torch.einsum("bcqhn,bckhn->bchqk",
-                          Ch.to(mm_dtype), Bh.to(mm_dtype))
+    scores_g = torch.einsum("bcqgn,bckgn->bcgqk", Cm, Bm)  # (b,nc,g,Q,Q)
+    sL = scores_g.view(b, nc, g, 1, chunk, chunk) * \
+        L.view(b, nc, g, rep, chunk, chunk)
     xdt = xc * dtc.unsqueeze(-1)                     # (b,nc,Q,h,p)
     xdt_m = xdt.to(mm_dtype)
-    y_diag = torch.einsum("bchqk,bckhp->bcqhp", scores * L, xdt_m).float()
+    y_diag = torch.einsum(
+        "bcgrqk,bckgrp->bcqgrp", sL,
+        xdt_m.view(b, nc, chunk, g, rep, p)).float() \
+        .reshape(b, nc, chunk, h, p)
 
     # chunk-final states: S_c = sum_j exp(dA_end - dA_j) B_j^T (dt_j x_j)
+    # — the decay is per-head, so fold it into xdt instead of
+    # materializing a decayed copy of B over all heads
     decay_states = torch.exp(dA_cs[..., -1:] - dA_cs)           # (b,nc,h,Q)
-    Bd = Bh * decay_states.permute(0, 1, 3, 2).unsqueeze(-1)    # (b,nc,Q,h,n)
-    states = torch.einsum("bckhn,bckhp->bchnp",
-                          Bd.to(mm_dtype), xdt_m).float()       # (b,nc,h,n,p)
+    xdt_dec = xdt_m * decay_states.permute(0, 1, 3, 2) \
+        .unsqueeze(-1).to(mm_dtype)
+    states = torch.einsum(
+        "bckgn,bckgrp->bcgrnp", Bm,
+        xdt_dec.view(b, nc, chunk, g, rep, p)).float() \
+        .reshape(b, nc, h, n, p)                                # (b,nc,h,n,p)
 
     # inter-chunk recurrence in closed form: prev[z] = sum_{c<z}
     # (prod_{c<k<z} D_k) S_c = (exp(segsum(log D)) @ S)[z-1] — one einsum
@@ -132,11 +141,15 @@ def ssd_chunked(x, dt, A, B, C, chunk):
     prev_states = torch.cat(
         [torch.zeros_like(P[:, :1]), P[:, :-1]], dim=1)         # (b,nc,h,n,p)
 
-    # off-diagonal: Y_off[i] = C_i exp(dA_cs[i]) S_{c-1}
+    # off-diagonal: Y_off[i] = C_i exp(dA_cs[i]) S_{c-1} — C stays per
+    # group; the per-(q,h) decay multiplies the GEMM OUTPUT (size h*p)
+    # instead of a materialized decayed C copy (size h*n)
     state_decay = torch.exp(dA_cs)                              # (b,nc,h,Q)
-    Cd = Ch * state_decay.permute(0, 1, 3, 2).unsqueeze(-1)     # (b,nc,Q,h,n)
-    y_off = torch.einsum("bcqhn,bchnp->bcqhp",
-                         Cd.to(mm_dtype), prev_states.to(mm_dtype)).float()
+    y_off = torch.einsum(
+        "bcqgn,bcgrnp->bcqgrp", Cm,
+        prev_states.view(b, nc, g, rep, n, p).to(mm_dtype)).float() \
+        .reshape(b, nc, chunk, h, p)
+    y_off = y_off * state_decay.permute(0, 1, 3, 2).unsqueeze(-1)
     return (y_diag + y_off).reshape(b, l, h, p)
 
 
