@@ -251,7 +251,7 @@ Tensor cconv_fwd(Tensor x, Tensor w, Tensor bias) {
   CHECK_BF16_CONTIG(w);
   const int b = x.size(0), l = x.size(1), c = x.size(2);
   const int W = w.size(1);
-  TORCH_CHECK(c % 8 == 0 && W <= 4);
+  TORCH_CHECK(c % 8 == 0 && W >= 2 && W <= 4);
   TORCH_CHECK(bias.scalar_type() == torch::kFloat32);
   auto y = torch::empty_like(x);
   launch_cconv_fwd(x.data_ptr(), w.data_ptr(), bias.data_ptr<float>(),
@@ -265,6 +265,7 @@ std::tuple<Tensor, Tensor, Tensor> cconv_bwd(Tensor dy, Tensor x, Tensor w,
   CHECK_BF16_CONTIG(x);
   const int b = x.size(0), l = x.size(1), c = x.size(2);
   const int W = w.size(1);
+  TORCH_CHECK(c % 8 == 0 && W >= 2 && W <= 4);
   auto g = torch::empty_like(x);
   auto dx = torch::empty_like(x);
   auto dw = torch::zeros({c, W}, x.options().dtype(torch::kFloat32));
