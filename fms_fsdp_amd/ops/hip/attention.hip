@@ -133,7 +133,6 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   constexpr int KVB = 64;
   constexpr int NC = D / 16;   // QK^T k-chunks
   constexpr int NT = D / 32;   // 32-wide output tiles
-  constexpr int DBLK = D / 16;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // double-buffered SUBTILED K/V images (same layout as the backward
   // kernels: coalesced b128 staging, plain reads for the S^T A-operand,
@@ -352,7 +351,6 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   constexpr int KVB = 32;
   constexpr int NC = D / 16;
   constexpr int NT = D / 32;
-  constexpr int DBLK = D / 16;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // double-buffered subtiled K/V images: stage tile t+1 during tile t's
   // compute, ONE barrier per tile.
@@ -515,7 +513,6 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   constexpr int KSWZ = (D == 128) ? 15 : 7;
   constexpr int NC = D / 16;
   constexpr int NT = D / 32;
-  constexpr int DBLK = D / 16;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                           // [4][KVB][D*2] per-wave
   // double-buffered subtiled Q/dO images (stage q-tile t+1 during t)
