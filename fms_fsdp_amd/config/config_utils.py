@@ -53,6 +53,9 @@ def get_model_config(model_variant):
         "llama2_13b": dict(emb_dim=5120, nheads=40, nlayers=40,
                            hidden_grow_factor=13824 / 5120),
         "llama2_7b": dict(hidden_grow_factor=11008 / 4096, kvheads=32),
+        # CPU-trivial entry for contract/smoke tests
+        "llama2_test": dict(src_vocab_size=256, emb_dim=64, nheads=2,
+                            kvheads=2, nlayers=2, max_expected_seq_len=128),
         "llama2_1.4b": dict(emb_dim=2048, nheads=16, nlayers=24,
                             hidden_grow_factor=3, kvheads=4),
         "llama3_8b": dict(src_vocab_size=128256, emb_dim=4096, nheads=32, kvheads=8,
