@@ -519,7 +519,12 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   constexpr int IMGB = 32 * D * 2;
 #define QIMG(buf) (smem + 4 * KVB * D * 2 + ((buf) ? 2 * IMGB : 0))
 #define DOIMG(buf) (smem + 4 * KVB * D * 2 + IMGB + ((buf) ? 2 * IMGB : 0))
-  char* p_lds = smem + 4 * KVB * D * 2 + 4 * IMGB;  // [4][KVB][32*2]
+  // P/dS transpose buffer: rows padded 64 -> 80 B so banks rotate 20
+  // per row — the 16-lane transposed reads land on 16 distinct 4-bank
+  // groups with no XOR (the old 64 B rows + 2-bit XOR were 4-way
+  // conflicted; dkv had 2x the bank conflicts of the other kernels,
+  // profiles/attn_pmc_counters_r01.md). 80 preserves 16 B alignment.
+  char* p_lds = smem + 4 * KVB * D * 2 + 4 * IMGB;  // [4][KVB][80]
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -541,7 +546,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   const short* vbase = vg + (long long)b * S * vstride + (long long)kvh * D;
 
   char* my_k = k_lds + wid * KVB * D * 2;
-  char* my_p = p_lds + wid * KVB * 64;
+  char* my_p = p_lds + wid * KVB * 80;
 
   // stage this wave's K rows once (swizzled row-major)
   {
@@ -632,7 +637,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key = DROW(r, hb);
-      *(short*)(my_p + ((key * 64 + col * 2) ^ ((key & 3) << 4))) = f2bf(pv[r]);
+      *(short*)(my_p + key * 80 + col * 2) = f2bf(pv[r]);
     }
     // dV[key][dv] += P(A) @ tr(imgdo)(B)
     {
@@ -640,7 +645,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         const int inrow = kc * 32 + hb * 16;
-        pa[kc] = *(const bf16x8v*)(my_p + ((col * 64 + inrow) ^ ((col & 3) << 4)));
+        pa[kc] = *(const bf16x8v*)(my_p + col * 80 + inrow);
       }
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
@@ -664,14 +669,14 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key = DROW(r, hb);
-      *(short*)(my_p + ((key * 64 + col * 2) ^ ((key & 3) << 4))) = f2bf(ds[r]);
+      *(short*)(my_p + key * 80 + col * 2) = f2bf(ds[r]);
     }
     {
       bf16x8v da[2];
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         const int inrow = kc * 32 + hb * 16;
-        da[kc] = *(const bf16x8v*)(my_p + ((col * 64 + inrow) ^ ((col & 3) << 4)));
+        da[kc] = *(const bf16x8v*)(my_p + col * 80 + inrow);
       }
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
@@ -759,7 +764,7 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
     attn_bwd_dq_kernel<128><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale, vstride);
-    const int lds_dkv = 4 * 32 * 128 * 2 + 4 * 32 * 128 * 2 + 4 * 32 * 64;
+    const int lds_dkv = 4 * 32 * 128 * 2 + 4 * 32 * 128 * 2 + 4 * 32 * 80;
     if (out_bf16)
       attn_bwd_dkv_kernel<128, true><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
@@ -775,7 +780,7 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
     attn_bwd_dq_kernel<64><<<grid, 256, lds_dq, stream>>>(
         (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
         lse, delta_ws, (short*)dq, B, S, H, KVH, scale, vstride);
-    const int lds_dkv = 4 * 32 * 64 * 2 + 4 * 32 * 64 * 2 + 4 * 32 * 64;
+    const int lds_dkv = 4 * 32 * 64 * 2 + 4 * 32 * 64 * 2 + 4 * 32 * 80;
     if (out_bf16)
       attn_bwd_dkv_kernel<64, true><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
