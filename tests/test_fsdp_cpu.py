@@ -97,6 +97,21 @@ def test_world2_matches_single_process(strat, reshard, port):
         assert abs(a - b) < 1e-5, (strat, ref, got)
 
 
+def test_world8_fsdp_matches_single_process():
+    """8-rank FSDP over gloo — the CPU proxy for the driver's single-node
+    8-GPU scaling run (one shard group of 8, the exact topology bench.py
+    sees at N=8). Trajectory must match a single process."""
+    ref = _single_process_reference()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_worker, args=(8, PORT + 30, "fsdp", False, q), nprocs=8,
+             join=True)
+    status, payload = q.get()
+    assert status == "ok", payload
+    for a, b in zip(ref, payload):
+        assert abs(a - b) < 1e-5, (ref, payload)
+
+
 def _ckpt_worker(rank, world, port, tmpdir, q):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
