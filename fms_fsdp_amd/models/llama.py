@@ -108,7 +108,12 @@ class RotaryEmbedding(nn.Module):
         return self.cos[:seqlen], self.sin[:seqlen]
 
     def reset_parameters(self):
-        pass
+        # rebuild the tables — required after a meta-device to_empty(),
+        # which leaves the buffers allocated but uninitialized
+        cos, sin = self._build(self.max_seq_len)
+        with torch.no_grad():
+            self.cos.copy_(cos)
+            self.sin.copy_(sin)
 
 
 class Attention(nn.Module):
@@ -295,10 +300,16 @@ class Llama(nn.Module):
             return tokens, torch.cat(embeds, dim=1)
         return tokens
 
-    def reset_parameters(self):
+    def reset_root_parameters(self):
+        """Init of the non-block params only — called alone by the
+        streamed meta-device path (ShardedModel._materialize_from_meta)
+        so the RNG draw order matches reset_parameters() exactly."""
         nn.init.trunc_normal_(self.embedding.weight, mean=0.0, std=0.02)
         nn.init.trunc_normal_(self.lm_head.weight, mean=0.0, std=0.02)
         self.norm.reset_parameters()
+
+    def reset_parameters(self):
+        self.reset_root_parameters()
         for layer in self.layers:
             layer.reset_parameters()
 

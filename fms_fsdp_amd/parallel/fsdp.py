@@ -52,7 +52,8 @@ class FlatUnit:
 
     def __init__(self, name: str, module: nn.Module, params: List[nn.Parameter],
                  shard_group, replicate_group, device, param_dtype,
-                 reshard_after_forward: bool, param_names=None):
+                 reshard_after_forward: bool, param_names=None,
+                 reduce_dtype=None):
         self.name = name
         self.module = module
         self.params = params
@@ -61,6 +62,7 @@ class FlatUnit:
         self.replicate_group = replicate_group    # None => no replication group
         self.device = device
         self.param_dtype = param_dtype
+        self.reduce_dtype = reduce_dtype or param_dtype
         self.reshard_after_forward = reshard_after_forward
         self.S = dist.get_world_size(shard_group) if shard_group is not None else 1
 
@@ -111,13 +113,15 @@ class FlatUnit:
             self.exp_avg = torch.zeros_like(self.master_shard)
             self.exp_avg_sq = torch.zeros_like(self.master_shard)
             self.flat_grad = torch.zeros(self.total, dtype=self.param_dtype, device=dev)
-            # grads stay in the param/reduce dtype end-to-end (bf16 reduce
-            # like the reference's bfSixteen policy); for S==1 the "shard"
-            # aliases flat_grad so the optimizer reads autograd's buffer
-            # with zero extra memory traffic.
-            if self.S > 1:
+            # grads accumulate in param_dtype (autograd writes flat_grad);
+            # the reduce-scatter output lives in reduce_dtype (reference
+            # MixedPrecision.reduce_dtype, mixed_precision.py:5-27). When
+            # S==1 and the dtypes match, the "shard" aliases flat_grad so
+            # the optimizer reads autograd's buffer with zero extra
+            # memory traffic.
+            if self.S > 1 or self.reduce_dtype != self.param_dtype:
                 self.grad_shard = torch.zeros(self.shard_size,
-                                              dtype=self.param_dtype, device=dev)
+                                              dtype=self.reduce_dtype, device=dev)
             else:
                 self.grad_shard = self.flat_grad
             self._point_params_into_flat()
@@ -154,6 +158,33 @@ class FlatUnit:
         st = self.flat_param.untyped_storage()
         if st.size() == 0:
             st.resize_(self.total * self.flat_param.element_size())
+
+    def free_flat_grad(self, rs_stream):
+        """Release the full-size grad buffer after its reduce-scatter has
+        been issued (reshard_after_forward memory regime: at 70B the flat
+        grads alone are 140 GB if kept resident — they must be transient).
+        The buffer is re-allocated and zeroed in prepare_grads() at the
+        unit's next backward-pre."""
+        if self.grad_shard is self.flat_grad:
+            return
+        if rs_stream is not None:
+            # the caching allocator must not hand this block out until
+            # the reduce-scatter reading it on rs_stream has passed
+            self.flat_grad.record_stream(rs_stream)
+        self.flat_grad.untyped_storage().resize_(0)
+
+    def prepare_grads(self):
+        """(Re-)allocate + zero the flat grad buffer if it was freed.
+        Zeroes everything (including alignment padding, which
+        reduce-scatter reads); direct-wgrad slices are marked fresh so
+        their first wgrad GEMM overwrites instead of accumulating."""
+        st = self.flat_grad.untyped_storage()
+        if st.size() == 0:
+            st.resize_(self.total * self.flat_grad.element_size())
+            self.flat_grad.zero_()
+            for p in self.params:
+                if getattr(p, "_direct_wgrad", False):
+                    p._wgrad_fresh = True
 
     # ---------------- collectives ----------------
 
@@ -195,9 +226,9 @@ class FlatUnit:
             else:
                 self._gathered = False  # keep storage, re-gather into it
 
-    def reduce_grads(self, rs_stream, reduce_dtype):
-        """reduce-scatter flat bf16 grads -> fp32 grad_shard (+= for grad
-        accumulation); HSDP: all-reduce the shard across replicas."""
+    def reduce_grads(self, rs_stream):
+        """reduce-scatter flat grads -> grad_shard in reduce_dtype; HSDP:
+        all-reduce the shard across replicas."""
         if rs_stream is not None:
             ev = torch.cuda.Event()
             ev.record(torch.cuda.current_stream())
@@ -207,8 +238,12 @@ class FlatUnit:
         ctx = torch.cuda.stream(rs_stream) if rs_stream is not None else nullcontext()
         with ctx:
             if self.S > 1:
-                dist.reduce_scatter_tensor(self.grad_shard, self.flat_grad,
+                src = self.flat_grad if self.flat_grad.dtype == self.reduce_dtype \
+                    else self.flat_grad.to(self.reduce_dtype)
+                dist.reduce_scatter_tensor(self.grad_shard, src,
                                            op=dist.ReduceOp.SUM, group=self.shard_group)
+            elif self.grad_shard is not self.flat_grad:
+                self.grad_shard.copy_(self.flat_grad)   # dtype cast
             if self.replicate_group is not None:
                 dist.all_reduce(self.grad_shard, group=self.replicate_group)
             if total_dp > 1:
@@ -295,7 +330,8 @@ class ShardedModel(nn.Module):
             block_param_ids.update(id(p) for p in params)
             self.units.append(FlatUnit(n, m, params, shard_group, replicate_group,
                                        device, param_dtype, reshard_after_forward,
-                                       param_names=[name_of[id(p)] for p in params]))
+                                       param_names=[name_of[id(p)] for p in params],
+                                       reduce_dtype=reduce_dtype))
         root_params = [p for p in model.parameters()
                        if p.requires_grad and id(p) not in block_param_ids]
         self.root_unit = None
@@ -303,24 +339,88 @@ class ShardedModel(nn.Module):
             self.root_unit = FlatUnit(
                 "_root", model, root_params, shard_group, replicate_group,
                 device, param_dtype, False,
-                param_names=[name_of[id(p)] for p in root_params])
+                param_names=[name_of[id(p)] for p in root_params],
+                reduce_dtype=reduce_dtype)
         self.all_units = ([self.root_unit] if self.root_unit else []) \
             + self.units
         self._unit_of_module = {id(u.module): u for u in self.units}
 
-        # move buffers (e.g. rope tables) to device, keep dtype
-        for b in model.buffers():
-            b.data = b.data.to(device)
-
-        # materialize shards (broadcast rank-0 init for determinism)
-        for u in self.all_units:
-            u.materialize(src_rank_broadcast=True)
+        if any(p.is_meta for p in model.parameters()):
+            # streamed materialization: one unit at a time from meta, so
+            # no rank ever holds the whole unsharded model (reference
+            # low_cpu_fsdp semantics, param_init.py:9-18; required for
+            # 70B: the full fp32 model would be ~280 GB per rank)
+            self._materialize_from_meta()
+        else:
+            # move buffers (e.g. rope tables) to device, keep dtype
+            for b in model.buffers():
+                b.data = b.data.to(device)
+            # materialize shards (broadcast rank-0 init for determinism)
+            for u in self.all_units:
+                u.materialize(src_rank_broadcast=True)
 
         # streams
         self.comm_stream = torch.cuda.Stream() if self.is_cuda else None
         self.rs_stream = torch.cuda.Stream() if self.is_cuda else None
 
         self._install_hooks()
+
+    def _materialize_from_meta(self):
+        """Materialize units one at a time from a meta-device model.
+
+        Per unit: to_empty() its owning modules on device, run the same
+        reset_parameters() the eager path uses (root unit:
+        model.reset_root_parameters()), copy into the unit's flat bf16
+        buffer (rank-0 broadcast), then swap param storage to flat-buffer
+        views so the fp32 originals are freed before the next unit is
+        touched. Peak transient memory = ONE unit's init params + its
+        flat buffer, never the whole model.
+        """
+        model, dev = self.model, self.device
+        # param id -> (owning module, local name); needed because
+        # to_empty() replaces the Parameter objects, so each unit's
+        # params list must be refreshed afterwards
+        slot_of = {}
+        for m in model.modules():
+            for n, p in m.named_parameters(recurse=False):
+                slot_of[id(p)] = (m, n)
+        # buffer-only modules first (rope tables): materialize + rebuild
+        for m in model.modules():
+            if any(b.is_meta for b in m.buffers(recurse=False)) and \
+                    not any(True for _ in m.parameters(recurse=False)):
+                m.to_empty(device=dev, recurse=False)
+                if hasattr(m, "reset_parameters"):
+                    m.reset_parameters()
+        for u in self.all_units:   # root first: same RNG order as eager
+            slots = [slot_of[id(p)] for p in u.params]
+            owners = []
+            for m, _ in slots:
+                if m not in owners:
+                    owners.append(m)
+            for m in owners:
+                m.to_empty(device=dev, recurse=False)
+            if u is self.root_unit and hasattr(model, "reset_root_parameters"):
+                model.reset_root_parameters()
+            elif u is not self.root_unit and hasattr(u.module, "reset_parameters"):
+                u.module.reset_parameters()
+            else:
+                for m in owners:
+                    if hasattr(m, "reset_parameters"):
+                        m.reset_parameters()
+            # refresh param refs (to_empty swapped the objects), carrying
+            # over the direct-wgrad routing flag set in module __init__
+            for (m, n), old in zip(slots, u.params):
+                if getattr(old, "_direct_wgrad", False):
+                    m._parameters[n]._direct_wgrad = True
+            u.params = [m._parameters[n] for m, n in slots]
+            u.materialize(src_rank_broadcast=True)
+        # any remaining non-unit buffers (none have params)
+        for b in model.buffers():
+            if b.is_meta:
+                raise RuntimeError("meta buffer left unmaterialized — give "
+                                   "its module a reset_parameters()")
+            if b.device != dev:
+                b.data = b.data.to(dev)
 
     # ---------------- hooks / orchestration ----------------
 
@@ -362,6 +462,7 @@ class ShardedModel(nn.Module):
     def _make_bwd_pre(self, idx):
         def hook(module, grad_output):
             u = self.units[idx]
+            u.prepare_grads()
             if not u._gathered:
                 u.gather(self.comm_stream)
             u.wait_gather()
@@ -379,9 +480,10 @@ class ShardedModel(nn.Module):
             u._grad_countdown = len(u.params)
         u._grad_countdown -= 1
         if u._grad_countdown == 0:
-            u.reduce_grads(self.rs_stream, self.reduce_dtype)
+            u.reduce_grads(self.rs_stream)
             if u.reshard_after_forward:
                 u._free_flat_param()
+                u.free_flat_grad(self.rs_stream)
 
     def _make_grad_hook(self, u: FlatUnit):
         def hook(param):
@@ -421,6 +523,10 @@ class ShardedModel(nn.Module):
     def zero_grad(self, set_to_none=False):
         self._clip_coef = None
         for u in self.all_units:
+            if u.flat_grad.untyped_storage().size() == 0:
+                # transient grads (reshard_after_forward): zeroed by
+                # prepare_grads() at the unit's backward-pre instead
+                continue
             # direct-wgrad slices are never pre-zeroed: their first
             # wgrad of the step overwrites (beta=0 addmm). Only the
             # ordinary-param slices (norm weights, embeddings) need the
@@ -436,7 +542,7 @@ class ShardedModel(nn.Module):
                         p._wgrad_fresh = True
                     else:
                         u.flat_grad[off:off + p.numel()].zero_()
-            if u.S > 1:
+            if u.grad_shard is not u.flat_grad:
                 u.grad_shard.zero_()
             if not u._grads_ready_views:
                 u._set_grad_views()

@@ -40,9 +40,7 @@ def get_latest(targdir, qualifier=lambda x: True, key="step"):
         full = os.path.join(targdir, fi)
         if key in fi and qualifier(full):
             try:
-                num = int("".join(c for c in fi.split(key)[1].split("_")[1]
-                                  if c.isdigit())) if False else \
-                    int(fi.replace("_ckp", "").split("_")[-1])
+                num = int(fi.replace("_ckp", "").split("_")[-1])
             except ValueError:
                 continue
             if num > best:
@@ -76,7 +74,10 @@ class Checkpointer:
         self.max_ckps = n_to_save
         self.rank = rank
         self.local_rank = local_rank
-        self.ckpt_dir = ckpt_dir
+        # all checkpoints live under a `checkpoints/` subfolder of the
+        # configured save dir (reference: checkpointing_utils.py:107),
+        # matching CheckpointDataset's loader-side layout
+        self.ckpt_dir = os.path.join(ckpt_dir, "checkpoints")
         assert parallel_mode in ("fsdp", "hsdp", "ddp")
         self.parallel_mode = parallel_mode
         self.report = report_fn if report_fn is not None else self._default_report
@@ -132,17 +133,41 @@ class Checkpointer:
             if self.rank == 0 and oldest != self.ckpt_dir:
                 shutil.rmtree(oldest, ignore_errors=True)
 
+    @staticmethod
+    def _is_complete(path):
+        """A checkpoint folder counts as valid only when metadata.pth exists
+        AND every model/optim shard it names finished writing (each shard
+        file gets a `.done` marker after its torch.save returns). This lets
+        auto-discovery skip a checkpoint truncated by a mid-save crash and
+        fall back to an older complete one."""
+        if os.path.isfile(path):
+            return True
+        mp = os.path.join(path, "metadata.pth")
+        if not os.path.exists(mp):
+            return False
+        try:
+            meta = torch.load(mp, map_location="cpu", weights_only=False)
+            S = meta["shard_world"]
+        except Exception:
+            return False
+        for r in range(S):
+            for kind in ("model", "optim"):
+                f = os.path.join(path, f"{kind}_{r}_of_{S}.pth")
+                if not (os.path.exists(f) and os.path.exists(f + ".done")):
+                    return False
+        return True
+
     def _validate_ckp_path(self, path):
-        """Return a valid checkpoint target under/at path, else None
-        (reference: checkpointing_utils.py:165-182)."""
+        """Return a valid (complete) checkpoint target under/at path, else
+        None (reference: checkpointing_utils.py:165-182)."""
         if not os.path.exists(path):
             return None
         if os.path.isfile(path):
             return path
-        if os.path.exists(os.path.join(path, "metadata.pth")):
+        if self._is_complete(path):
             return path
-        latest = get_latest(path)
-        if latest != path and os.path.exists(os.path.join(latest, "metadata.pth")):
+        latest = get_latest(path, qualifier=self._is_complete)
+        if latest != path and self._is_complete(latest):
             return latest
         if os.path.isfile(latest):
             return latest
@@ -160,6 +185,12 @@ class Checkpointer:
             dist.barrier()
         # D2H snapshots first (before training mutates the shards), then
         # serialize — in a background thread when async_save is on.
+        # NOTE on loader-state fidelity: dataloader.dataset.state_dict()
+        # here snapshots the MAIN-process copy; with num_workers>0 the live
+        # iteration state is in the DataLoader worker, so these files lag
+        # by up to the prefetch depth. Token-exact resume relies on
+        # CheckpointDataset's in-worker saves (data/datasets.py); set
+        # num_workers=0 if exact resume through Checkpointer alone matters.
         payloads = []
         if is_writer:
             model_sd = {u.name: u.master_shard.cpu() for u in model.all_units}
@@ -174,25 +205,39 @@ class Checkpointer:
         if dataloader is not None and hasattr(dataloader.dataset, "state_dict"):
             payloads.append((dataloader.dataset.state_dict(),
                              os.path.join(out, f"loader_state_{self.rank}.pth")))
-        if self.rank == 0:
-            # metadata last: it is the validity marker load() looks for
-            payloads.append(({"step": step, "tokens_seen": tokens_seen,
-                              "shard_world": S, "layout": self._layout(model)},
-                             os.path.join(out, "metadata.pth")))
+        meta = ({"step": step, "tokens_seen": tokens_seen,
+                 "shard_world": S, "layout": self._layout(model)}
+                if self.rank == 0 else None)
         if self.async_save:
+            # No cross-rank barrier is possible from the writer thread
+            # (training is issuing its own collectives); checkpoint
+            # validity instead comes from the per-shard .done markers that
+            # _is_complete() requires from EVERY shard rank, so metadata
+            # appearing early never makes a partial checkpoint loadable.
             self._pending = threading.Thread(
-                target=self._write_payloads, args=(payloads, out, t0),
+                target=self._write_payloads, args=(payloads, meta, out, t0),
                 daemon=True)
             self._pending.start()
         else:
-            self._write_payloads(payloads, out, t0)
+            self._write_payloads(payloads, None, out, t0)
+            if dist.is_initialized():
+                # all shard writes land before rank 0 publishes metadata
+                dist.barrier()
+            if meta is not None:
+                torch.save(meta, os.path.join(out, "metadata.pth"))
             if dist.is_initialized():
                 dist.barrier()
         return out
 
-    def _write_payloads(self, payloads, out, t0):
+    def _write_payloads(self, payloads, meta, out, t0):
         for obj, path in payloads:
             torch.save(obj, path)
+            base = os.path.basename(path)
+            if base.startswith(("model_", "optim_")):
+                with open(path + ".done", "w"):
+                    pass
+        if meta is not None:
+            torch.save(meta, os.path.join(out, "metadata.pth"))
         self.report(output_path=out, time_taken=f"{time.time() - t0:.2f}s")
         self._cleanup()
 
@@ -238,11 +283,13 @@ class Checkpointer:
                                     shard_rank, S, old_S)
         is_resuming = not reset_stepcount
         if dataloader is not None and is_resuming and \
-                hasattr(dataloader.dataset, "load_state_dict"):
-            lp = os.path.join(load_path, f"loader_state_{self.rank}.pth")
-            if os.path.exists(lp):
-                dataloader.dataset.load_state_dict(
-                    torch.load(lp, map_location="cpu", weights_only=False))
+                hasattr(dataloader.dataset, "load_from_path"):
+            # load_from_path reads ALL loader_state_<r>.pth files and
+            # reshards when the world size changed (the _StatefulDataset
+            # API takes a list of per-rank states, not one flat dict —
+            # reference: checkpointing_utils.py:274-278).
+            if any(f.startswith("loader_state_") for f in os.listdir(load_path)):
+                dataloader.dataset.load_from_path(load_path)
         self.report(msg=f"Checkpoint loaded from {load_path}", step=step)
         return model, optimizer, dataloader, step, tokens, is_resuming
 

@@ -77,11 +77,12 @@ def main(**kwargs):
     model_config = get_model_config(cfg.model_variant)
     model_config.src_vocab_size = max(model_config.src_vocab_size, cfg.vocab_size)
     if cfg.low_cpu_fsdp:
+        # model stays on meta; ShardedModel streams materialization one
+        # unit at a time (no rank ever holds the full unsharded model —
+        # reference: param_init.py:9-18 + main_training_llama.py:59-65)
         with torch.device("meta"):
             model = Llama(model_config)
-        model = model.to_empty(device="cuda" if torch.cuda.is_available() else "cpu")
         torch.manual_seed(cfg.seed)  # identical init on every rank pre-broadcast
-        model.reset_parameters()
     else:
         model = Llama(model_config)
         model.reset_parameters()

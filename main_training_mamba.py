@@ -74,7 +74,8 @@ def main(**kwargs):
     model, optimizer, _, start_step, tokens_seen, _ = checkpointer.load(
         model, optimizer,
         None if cfg.use_dummy_dataset else train_loader,
-        path=cfg.ckpt_load_path)
+        path=os.path.join(cfg.ckpt_load_path, "checkpoints/")
+        if not os.path.isfile(cfg.ckpt_load_path) else cfg.ckpt_load_path)
 
     warmup_interval = min(2000, cfg.num_steps // 20) or 1
     schedule = lambda x: min(

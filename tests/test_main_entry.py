@@ -16,7 +16,7 @@ def test_main_llama_smoke(tmp_path):
         ckpt_save_path=str(tmp_path), ckpt_load_path=str(tmp_path),
         vocab_size=256, learning_rate=1e-4, sharding_strategy="fsdp")
     # a checkpoint was written at step 2
-    assert os.path.exists(tmp_path / "step_2_ckp" / "metadata.pth")
+    assert os.path.exists(tmp_path / "checkpoints" / "step_2_ckp" / "metadata.pth")
     # resume: runs steps 3.. from the checkpoint
     main_training_llama.main(
         model_variant="llama2_125m", use_dummy_dataset=True, batch_size=1,
@@ -24,7 +24,7 @@ def test_main_llama_smoke(tmp_path):
         checkpoint_interval=2, mixed_precision=False,
         ckpt_save_path=str(tmp_path), ckpt_load_path=str(tmp_path),
         vocab_size=256, learning_rate=1e-4, sharding_strategy="fsdp")
-    assert os.path.exists(tmp_path / "step_4_ckp" / "metadata.pth")
+    assert os.path.exists(tmp_path / "checkpoints" / "step_4_ckp" / "metadata.pth")
 
 
 def test_main_llama_selective_ac(tmp_path):
