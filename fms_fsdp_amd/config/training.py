@@ -92,5 +92,15 @@ class train_config:
     reshard_after_forward: Union[bool, str] = "auto"
     # How many units ahead to prefetch all-gathers on the comm stream.
     prefetch_lookahead: int = 1
-    # low-precision reduce-scatter of grads (bf16) with fp32 master accum
-    reduce_dtype: str = "bf16"
+    # low-precision reduce-scatter of grads (bf16) with fp32 master accum;
+    # "policy" defers to the mp_policy triple below
+    reduce_dtype: str = "policy"
+    # Mixed-precision policy (reference mixed_precision.py:5-27 triples):
+    #   auto         -> bf16 when mixed_precision and the device supports
+    #                   it, fp16 otherwise; fp32 when not mixed_precision
+    #   bf16         -> bfSixteen        (param bf16, reduce bf16)
+    #   bf16_working -> bfSixteen_working (param fp32, reduce bf16)
+    #   fp16         -> fpSixteen        (param fp16, reduce fp16,
+    #                   dynamic loss scaling enabled automatically)
+    #   fp32         -> fp32_policy
+    mp_policy: str = "auto"

@@ -13,6 +13,12 @@ import torch
 
 from . import reference
 
+# fp16 policy note (reference fpSixteen, mixed_precision.py:5-9): fp16 is
+# supported as the STORAGE/COMM dtype; op-internal compute bridges through
+# bf16 (identical MFMA rate on CDNA4, wider exponent, fp32 accumulation
+# inside every kernel) so the HIP kernels stay single-dtype. The casts are
+# autograd-tracked, so gradients flow back to the fp16 tensors.
+
 _C = None
 _C_err = None
 try:
@@ -54,6 +60,9 @@ class _RMSNormFn(torch.autograd.Function):
 
 def rmsnorm(x, weight, eps=1e-6):
     if x.is_cuda:
+        if x.dtype == torch.float16:
+            return _RMSNormFn.apply(x.bfloat16(), weight.bfloat16(),
+                                    eps).half()
         return _RMSNormFn.apply(x, weight, eps)
     return reference.rmsnorm(x, weight, eps)
 
@@ -84,6 +93,10 @@ class _AddRMSNormFn(torch.autograd.Function):
 def add_rmsnorm(x, res, weight, eps=1e-6):
     """(rmsnorm(x + res), x + res)"""
     if x.is_cuda:
+        if x.dtype == torch.float16:
+            y, s = _AddRMSNormFn.apply(x.bfloat16(), res.bfloat16(),
+                                       weight.bfloat16(), eps)
+            return y.half(), s.half()
         return _AddRMSNormFn.apply(x, res, weight, eps)
     s = (x.float() + res.float()).to(x.dtype)
     return reference.rmsnorm(s, weight, eps), s
@@ -127,6 +140,9 @@ class _RoPEFn(torch.autograd.Function):
 
 def rope_apply(q, k, cos, sin):
     if q.is_cuda:
+        if q.dtype == torch.float16:
+            qo, ko = _RoPEFn.apply(q.bfloat16(), k.bfloat16(), cos, sin)
+            return qo.half(), ko.half()
         return _RoPEFn.apply(q, k, cos, sin)
     return reference.rope_apply(q, k, cos, sin)
 
@@ -296,6 +312,9 @@ def qkv_rope_attention(qkv, cos, sin, nheads, kvheads, head_dim):
     """Fused GPU path; callers gate on seq%128==0 and head_dim in
     (64, 128) and fall back to the modular split+rope+attention path
     otherwise (decode/KV-cache, odd lengths, CPU)."""
+    if qkv.dtype == torch.float16:
+        return _QKVRopeAttnFn.apply(qkv.bfloat16(), cos, sin, nheads,
+                                    kvheads, head_dim).half()
     return _QKVRopeAttnFn.apply(qkv, cos, sin, nheads, kvheads, head_dim)
 
 
@@ -307,6 +326,9 @@ def attention_causal(q, k, v):
                 q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
                 is_causal=True, enable_gqa=(k.shape[2] != q.shape[2]))
             return o.transpose(1, 2)
+        if q.dtype == torch.float16:
+            return _FlashAttnFn.apply(q.bfloat16(), k.bfloat16(),
+                                      v.bfloat16()).half()
         return _FlashAttnFn.apply(q, k, v)
     return reference.attention_causal(q, k, v)
 
@@ -333,6 +355,8 @@ class _SwiGLUFn(torch.autograd.Function):
 
 def swiglu(gu):
     if gu.is_cuda:
+        if gu.dtype == torch.float16:
+            return _SwiGLUFn.apply(gu.bfloat16()).half()
         return _SwiGLUFn.apply(gu)
     return reference.swiglu(gu)
 
@@ -387,6 +411,9 @@ class _LinearCEFn(torch.autograd.Function):
 
 def linear_cross_entropy(x, weight, labels, ignore_index=-100):
     if x.is_cuda:
+        if x.dtype == torch.float16:
+            return _LinearCEFn.apply(x.bfloat16(), weight.bfloat16(),
+                                     labels, ignore_index)
         return _LinearCEFn.apply(x, weight, labels, ignore_index)
     return reference.linear_cross_entropy(x, weight, labels, ignore_index)
 
@@ -453,9 +480,10 @@ def segsum_exp(cs):
 # --------------------------------------------------------------------------
 def fused_adamw(p, g, m, v, step, lr, beta1, beta2, eps, weight_decay,
                 grad_scale=None, p_bf16_out=None):
-    """In-place AdamW on 1-D fp32 tensors. g may be fp32 or bf16;
-    grad_scale (0-dim fp32 tensor) folds grad clipping into the update;
-    p_bf16_out receives the updated bf16 shard in the same pass."""
+    """In-place AdamW on 1-D fp32 tensors. g may be fp32, bf16 or fp16;
+    grad_scale (0-dim fp32 tensor) folds grad clipping (and fp16 loss
+    unscaling) into the update; p_bf16_out receives the updated bf16 OR
+    fp16 shard in the same pass."""
     if p.is_cuda:
         ext = _require_ext("adamw")
         ext.adamw(p, g.view(-1), m, v, float(step), lr, beta1, beta2, eps,

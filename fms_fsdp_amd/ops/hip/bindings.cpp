@@ -22,7 +22,7 @@ void launch_swiglu_bwd(const void*, const void*, void*, long long, int,
                        hipStream_t);
 void launch_ce_fwd_bwd(void*, const long long*, float*, const float*,
                        long long, int, int, hipStream_t);
-void launch_adamw(float*, const void*, int, float*, float*, void*,
+void launch_adamw(float*, const void*, int, float*, float*, void*, int,
                   long long, float, float, float, float, float, float, float,
                   const float*, hipStream_t);
 void launch_sqnorm(const void*, int, float*, long long, hipStream_t);
@@ -176,12 +176,24 @@ void adamw(Tensor p, Tensor g, Tensor m, Tensor v, double step, double lr,
   TORCH_CHECK(p.scalar_type() == torch::kFloat32 && p.is_contiguous());
   const long long n = p.numel();
   TORCH_CHECK(n % 4 == 0, "shard size must be divisible by 4");
-  const bool gbf = g.scalar_type() == torch::kBFloat16;
+  auto dt_code = [](torch::ScalarType t) {
+    if (t == torch::kBFloat16) return 1;
+    if (t == torch::kFloat16) return 2;
+    TORCH_CHECK(t == torch::kFloat32, "dtype must be fp32/bf16/fp16");
+    return 0;
+  };
+  const int gdt = dt_code(g.scalar_type());
+  int odt = 0;
+  if (p_bf16_out.has_value()) {
+    odt = dt_code(p_bf16_out->scalar_type());
+    TORCH_CHECK(odt != 0, "low-precision publish target must be bf16/fp16");
+  }
   const float bc1 = 1.f - powf((float)b1, (float)step);
   const float bc2 = 1.f - powf((float)b2, (float)step);
-  launch_adamw(p.data_ptr<float>(), g.data_ptr(), gbf, m.data_ptr<float>(),
+  launch_adamw(p.data_ptr<float>(), g.data_ptr(), gdt, m.data_ptr<float>(),
                v.data_ptr<float>(),
-               p_bf16_out.has_value() ? p_bf16_out->data_ptr() : nullptr, n,
+               p_bf16_out.has_value() ? p_bf16_out->data_ptr() : nullptr, odt,
+               n,
                (float)lr, (float)b1, (float)b2, (float)eps, (float)wd, bc1,
                bc2,
                grad_scale.has_value() ? grad_scale->data_ptr<float>() : nullptr,
@@ -192,8 +204,9 @@ void sq_norm_accum(Tensor t, Tensor out) {
   TORCH_CHECK(t.is_contiguous());
   const long long n = t.numel();
   TORCH_CHECK(n % 4 == 0);
-  launch_sqnorm(t.data_ptr(), t.scalar_type() == torch::kBFloat16,
-                out.data_ptr<float>(), n, cur_stream());
+  const int dt = t.scalar_type() == torch::kBFloat16 ? 1
+                 : t.scalar_type() == torch::kFloat16 ? 2 : 0;
+  launch_sqnorm(t.data_ptr(), dt, out.data_ptr<float>(), n, cur_stream());
 }
 
 std::tuple<Tensor, Tensor> attn_fwd(Tensor q, Tensor k, Tensor v) {

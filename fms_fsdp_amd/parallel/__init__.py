@@ -1,3 +1,3 @@
-from .fsdp import ShardedModel, ShardedAdamW, FlatUnit
+from .fsdp import ShardedModel, ShardedAdamW, FlatUnit, DynamicGradScaler
 
-__all__ = ["ShardedModel", "ShardedAdamW", "FlatUnit"]
+__all__ = ["ShardedModel", "ShardedAdamW", "FlatUnit", "DynamicGradScaler"]

@@ -574,13 +574,13 @@ class ShardedAdamW:
         gscale = getattr(self.m, "_clip_coef", None)
         for u in self.m.all_units:
             u.wait_grads()
+            lowp = u.param_shard.dtype in (torch.bfloat16, torch.float16)
             published = ops.fused_adamw(
                 u.master_shard, u.grad_shard, u.exp_avg, u.exp_avg_sq,
                 self.step_count, lr, b1, b2, self.eps, self.weight_decay,
                 grad_scale=gscale,
-                p_bf16_out=u.param_shard
-                if u.param_shard.dtype == torch.bfloat16 else None)
-            if not published or u.param_shard.dtype != torch.bfloat16:
+                p_bf16_out=u.param_shard if lowp else None)
+            if not published or not lowp:
                 u.publish_master_to_shard()
             u.mark_stale()
 
@@ -606,6 +606,64 @@ class ShardedAdamW:
             u.master_shard.copy_(usd["master"])
             u.publish_master_to_shard()
             u.mark_stale()
+
+
+class DynamicGradScaler:
+    """Dynamic loss scaling for the fp16 policy (reference fpSixteen,
+    mixed_precision.py:5-9; torch.cuda.amp.GradScaler semantics).
+
+    Usage inside the train loop:
+        scaler.scale_loss(loss).backward()
+        gnorm, stepped = scaler.clip_and_step(model, optimizer, max_norm)
+
+    The unscale (1/scale) is FOLDED into the clip coefficient the fused
+    AdamW kernel applies, so no extra pass over the grads is ever made.
+    On inf/nan grad norm the step is skipped and the scale backs off;
+    after growth_interval consecutive good steps it doubles.
+    """
+
+    def __init__(self, enabled=True, init_scale=2.0 ** 16, growth_factor=2.0,
+                 backoff_factor=0.5, growth_interval=1000):
+        self.enabled = enabled
+        self.scale = float(init_scale) if enabled else 1.0
+        self.growth_factor = growth_factor
+        self.backoff_factor = backoff_factor
+        self.growth_interval = growth_interval
+        self._good_steps = 0
+
+    def scale_loss(self, loss):
+        return loss * self.scale if self.enabled else loss
+
+    def clip_and_step(self, model, optimizer, max_norm):
+        """Returns (true grad norm, stepped)."""
+        if not self.enabled:
+            gnorm = model.clip_grad_norm_(max_norm)
+            optimizer.step()
+            return gnorm, True
+        # clip computed on the SCALED grads against max_norm*scale:
+        # coef = clamp(max_norm*scale / norm_s, 1) then /scale gives
+        # exactly clamp(max_norm / true_norm, 1/scale) = clip ∘ unscale
+        gnorm_scaled = model.clip_grad_norm_(max_norm * self.scale)
+        true_norm = gnorm_scaled / self.scale
+        if not bool(torch.isfinite(gnorm_scaled)):
+            self.scale = max(self.scale * self.backoff_factor, 1.0)
+            self._good_steps = 0
+            return true_norm, False   # skip the update entirely
+        model._clip_coef = model._clip_coef / self.scale
+        optimizer.step()
+        self._good_steps += 1
+        if self._good_steps % self.growth_interval == 0:
+            self.scale *= self.growth_factor
+        return true_norm, True
+
+    def state_dict(self):
+        return {"scale": self.scale, "good_steps": self._good_steps,
+                "enabled": self.enabled}
+
+    def load_state_dict(self, sd):
+        self.scale = sd["scale"]
+        self._good_steps = sd["good_steps"]
+        self.enabled = sd["enabled"]
 
 
 def _build_hsdp_groups(world, rank, intra):

@@ -42,14 +42,47 @@ def apply_selective_ac(model, block_class, p):
     return pattern
 
 
+# (param_dtype, reduce_dtype) triples — reference mixed_precision.py:5-27
+# (buffer dtype is moot here: our buffers are the fp32 rope tables, which
+# the RoPE kernel reads in fp32 regardless)
+_MP_POLICIES = {
+    "fp16": (torch.float16, torch.float16),          # fpSixteen
+    "bf16": (torch.bfloat16, torch.bfloat16),        # bfSixteen
+    "bf16_working": (torch.float32, torch.bfloat16), # bfSixteen_working
+    "fp32": (torch.float32, torch.float32),          # fp32_policy
+}
+
+
+def resolve_mp_policy(cfg):
+    """cfg -> policy name (reference get_mixed_precision_policy,
+    train_utils.py:192-214: bf16 when supported, else fp16 fallback)."""
+    pol = (getattr(cfg, "mp_policy", "auto") or "auto").lower()
+    if pol == "auto":
+        if cfg.mixed_precision:
+            pol = "bf16" if _bf16_ready() else "fp16"
+        else:
+            pol = "fp32"
+    if pol not in _MP_POLICIES:
+        raise ValueError(f"unknown mp_policy {pol} "
+                         f"(one of auto|{'|'.join(_MP_POLICIES)})")
+    return pol
+
+
 def get_mixed_precision_dtypes(cfg):
-    """cfg -> (param_dtype, reduce_dtype) (reference mixed_precision.py:
-    bfSixteen = bf16/bf16/bf16; fp32_policy = fp32 everywhere)."""
-    if cfg.mixed_precision:
-        if not _bf16_ready():
-            raise ValueError("bf16 requested but not supported on this device/backend")
-        return torch.bfloat16, torch.bfloat16
-    return torch.float32, torch.float32
+    """cfg -> (param_dtype, reduce_dtype). cfg.reduce_dtype overrides the
+    policy's reduce dtype when set to an explicit dtype name."""
+    param_dtype, reduce_dtype = _MP_POLICIES[resolve_mp_policy(cfg)]
+    override = (getattr(cfg, "reduce_dtype", "policy") or "policy").lower()
+    names = {"bf16": torch.bfloat16, "fp16": torch.float16,
+             "fp32": torch.float32}
+    if override in names:
+        reduce_dtype = names[override]
+    return param_dtype, reduce_dtype
+
+
+def needs_loss_scaling(cfg):
+    """fp16 params can under/overflow without scaling; bf16/fp32 cannot."""
+    return resolve_mp_policy(cfg) == "fp16"
 
 
 def _bf16_ready():

@@ -112,8 +112,14 @@ def get_tracker(cfg, rank):
 
 def train(cfg, model, local_rank, rank, train_loader, optimizer, scheduler,
           profiler=None, checkpointer=None, start_step=0, n_tok=0,
-          tracker=None):
-    """Steady-state training loop (reference: train_utils.py:21-180)."""
+          tracker=None, scaler=None):
+    """Steady-state training loop (reference: train_utils.py:21-180).
+    `scaler`: DynamicGradScaler — required for the fp16 policy, a no-op
+    pass-through otherwise (built automatically from cfg when None)."""
+    from fms_fsdp_amd.parallel import DynamicGradScaler
+    from fms_fsdp_amd.parallel.policies import needs_loss_scaling
+    if scaler is None:
+        scaler = DynamicGradScaler(enabled=needs_loss_scaling(cfg))
     model.train()
     world = dist.get_world_size() if dist.is_initialized() else 1
     device = torch.device("cuda", local_rank) if torch.cuda.is_available() \
@@ -133,12 +139,14 @@ def train(cfg, model, local_rank, rank, train_loader, optimizer, scheduler,
 
         optimizer.zero_grad()
         loss = model(inp, labels=label)
-        loss.backward()
+        scaler.scale_loss(loss).backward()
         ddp_stats[0] += loss.detach().float()
-        gnorm = model.clip_grad_norm_(cfg.grad_clip_thresh)
-        ddp_stats[1] += gnorm.detach().float()
+        gnorm, _ = scaler.clip_and_step(model, optimizer,
+                                        cfg.grad_clip_thresh)
+        # (short-circuit keeps the default path free of host syncs)
+        if not scaler.enabled or bool(torch.isfinite(gnorm)):
+            ddp_stats[1] += gnorm.detach().float()
         ddp_stats[2] += 1
-        optimizer.step()
         scheduler.step()
 
         if profiler:

@@ -22,10 +22,21 @@ import torch.distributed as dist
 import torch.nn.functional as F
 
 
-def stage1_loss(cfg, model, speculator, base_model_input, inp, ddp_stats):
+def _tp_chunk(t, tp_group):
+    """TP mode runs the frozen base on the all-gathered (tp*b, ...) batch;
+    each rank trains its speculator only on ITS b-sized slice (reference
+    utils:158-162, 224-232 — omitting this mismatches the batch dims)."""
+    if tp_group is None or dist.get_world_size(tp_group) == 1:
+        return t
+    return t.chunk(dist.get_world_size(tp_group))[dist.get_rank(tp_group)]
+
+
+def stage1_loss(cfg, model, speculator, base_model_input, inp, ddp_stats,
+                tp_group=None):
     with torch.no_grad():
         _, embeds = model(base_model_input[:, : -speculator.n_predict - 1],
                           include_embeds=True)
+        embeds = _tp_chunk(embeds, tp_group)
     preds = speculator(embeds.detach(), inp[:, 1:])
     losses = []
     for i in range(preds.size(0)):
@@ -37,7 +48,8 @@ def stage1_loss(cfg, model, speculator, base_model_input, inp, ddp_stats):
     return sum(losses), ddp_stats, inp.numel()
 
 
-def stage2_loss(cfg, model, speculator, base_model_input, inp, ddp_stats):
+def stage2_loss(cfg, model, speculator, base_model_input, inp, ddp_stats,
+                tp_group=None):
     n = speculator.n_predict
     with torch.no_grad():
         grow = cfg.stage2_batch_size // cfg.batch_size
@@ -47,6 +59,8 @@ def stage2_loss(cfg, model, speculator, base_model_input, inp, ddp_stats):
             .reshape(base_model_input.size(0) * grow, cfg.stage2_prompt_length)
         targs, embeds = model.generate(prompts, cfg.stage2_seq_length,
                                        do_sample=True, include_embeds=True)
+        targs = _tp_chunk(targs, tp_group)
+        embeds = _tp_chunk(embeds, tp_group)
         gen = targs[:, -cfg.stage2_seq_length:]           # generated tokens
         state = embeds[:, : cfg.stage2_seq_length - n]
     preds = speculator(state.detach(), gen[:, :-1].detach())
@@ -101,10 +115,10 @@ def train_speculator(cfg, model, speculator, local_rank, rank, world_size,
         optimizer.zero_grad()
         if batch_idx <= cfg.stage2_start_step:
             loss, ddp_stats, step_tok = stage1_loss(
-                cfg, model, speculator, bmi, inp, ddp_stats)
+                cfg, model, speculator, bmi, inp, ddp_stats, tp_group)
         else:
             loss, ddp_stats, step_tok = stage2_loss(
-                cfg, model, speculator, bmi, inp, ddp_stats)
+                cfg, model, speculator, bmi, inp, ddp_stats, tp_group)
         loss.backward()
         ddp_stats[0] += speculator.clip_grad_norm_(cfg.grad_clip_thresh) \
             if hasattr(speculator, "clip_grad_norm_") else \

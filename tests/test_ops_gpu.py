@@ -469,3 +469,40 @@ def test_attention_padded_seq():
     assert relerr(o, ref) < 3e-2
     o.sum().backward()
     assert torch.isfinite(q.grad).all() and torch.isfinite(k.grad).all()
+
+
+@pytest.mark.parametrize("policy", ["fp16", "bf16_working"])
+def test_model_gpu_step_mp_policies(policy):
+    """fp16 (with dynamic loss scaling) and bf16_working policies on GPU:
+    loss decreases, fused AdamW handles the policy's grad/publish dtypes
+    (reference mixed_precision.py:5-27)."""
+    from fms_fsdp_amd.config import train_config
+    from fms_fsdp_amd.models import Llama, LlamaBlock, LlamaConfig
+    from fms_fsdp_amd.parallel import (DynamicGradScaler, ShardedAdamW,
+                                       ShardedModel)
+    from fms_fsdp_amd.parallel.policies import (get_mixed_precision_dtypes,
+                                                needs_loss_scaling)
+    torch.manual_seed(0)
+    cfg = LlamaConfig(src_vocab_size=512, emb_dim=256, nheads=2, kvheads=2,
+                      nlayers=2, max_expected_seq_len=256)
+    with torch.device(dev()):
+        m = Llama(cfg)
+        m.reset_parameters()
+    tc = train_config()
+    tc.mp_policy = policy
+    pd, rd = get_mixed_precision_dtypes(tc)
+    sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                      param_dtype=pd, reduce_dtype=rd)
+    opt = ShardedAdamW(sm, lr=1e-3)
+    scaler = DynamicGradScaler(enabled=needs_loss_scaling(tc))
+    x = torch.randint(0, 512, (2, 256), device=dev())
+    y = torch.randint(0, 512, (2, 256), device=dev())
+    losses = []
+    for _ in range(5):
+        opt.zero_grad()
+        loss = sm(x, labels=y)
+        scaler.scale_loss(loss).backward()
+        scaler.clip_and_step(sm, opt, 1.0)
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+    assert all(l == l for l in losses), losses
