@@ -78,9 +78,13 @@ def main():
         block_cls = LlamaBlock
     n_params = model.param_count()
 
+    from fms_fsdp_amd.config import train_config
+    from fms_fsdp_amd.parallel.policies import resolve_reshard_after_forward
+    _rcfg = train_config()   # "auto" resolver, same path the entries use
     sm = ShardedModel(model, block_cls, sharding_strategy=args.sharding,
                       param_dtype=torch.bfloat16 if use_cuda else torch.float32,
-                      reshard_after_forward=n_params > 30e9,
+                      reshard_after_forward=resolve_reshard_after_forward(
+                          _rcfg, n_params),
                       prefetch_lookahead=1, device=device)
     if args.ac not in ("0", "0.0", 0):
         apply_selective_ac(sm, block_cls, args.ac)

@@ -18,6 +18,12 @@ from . import reference
 # bf16 (identical MFMA rate on CDNA4, wider exponent, fp32 accumulation
 # inside every kernel) so the HIP kernels stay single-dtype. The casts are
 # autograd-tracked, so gradients flow back to the fp16 tensors.
+#
+# fp32 activations on GPU (bf16_working / fp32 policies) take the fp32
+# reference implementations by DESIGN: "working precision fp32" means the
+# math genuinely runs in fp32 (the bf16 MFMA kernels would silently drop
+# precision), and at fp32 rates the eager ops are not the bottleneck. The
+# default (bf16) path never touches this branch.
 
 _C = None
 _C_err = None
@@ -59,7 +65,7 @@ class _RMSNormFn(torch.autograd.Function):
 
 
 def rmsnorm(x, weight, eps=1e-6):
-    if x.is_cuda:
+    if x.is_cuda and x.dtype != torch.float32:
         if x.dtype == torch.float16:
             return _RMSNormFn.apply(x.bfloat16(), weight.bfloat16(),
                                     eps).half()
@@ -92,7 +98,7 @@ class _AddRMSNormFn(torch.autograd.Function):
 
 def add_rmsnorm(x, res, weight, eps=1e-6):
     """(rmsnorm(x + res), x + res)"""
-    if x.is_cuda:
+    if x.is_cuda and x.dtype != torch.float32:
         if x.dtype == torch.float16:
             y, s = _AddRMSNormFn.apply(x.bfloat16(), res.bfloat16(),
                                        weight.bfloat16(), eps)
@@ -139,7 +145,7 @@ class _RoPEFn(torch.autograd.Function):
 
 
 def rope_apply(q, k, cos, sin):
-    if q.is_cuda:
+    if q.is_cuda and q.dtype != torch.float32:
         if q.dtype == torch.float16:
             qo, ko = _RoPEFn.apply(q.bfloat16(), k.bfloat16(), cos, sin)
             return qo.half(), ko.half()
@@ -329,6 +335,8 @@ def attention_causal(q, k, v):
         if q.dtype == torch.float16:
             return _FlashAttnFn.apply(q.bfloat16(), k.bfloat16(),
                                       v.bfloat16()).half()
+        if q.dtype == torch.float32:
+            return reference.attention_causal(q, k, v)
         return _FlashAttnFn.apply(q, k, v)
     return reference.attention_causal(q, k, v)
 
@@ -354,7 +362,7 @@ class _SwiGLUFn(torch.autograd.Function):
 
 
 def swiglu(gu):
-    if gu.is_cuda:
+    if gu.is_cuda and gu.dtype != torch.float32:
         if gu.dtype == torch.float16:
             return _SwiGLUFn.apply(gu.bfloat16()).half()
         return _SwiGLUFn.apply(gu)
@@ -410,7 +418,7 @@ class _LinearCEFn(torch.autograd.Function):
 
 
 def linear_cross_entropy(x, weight, labels, ignore_index=-100):
-    if x.is_cuda:
+    if x.is_cuda and x.dtype != torch.float32:
         if x.dtype == torch.float16:
             return _LinearCEFn.apply(x.bfloat16(), weight.bfloat16(),
                                      labels, ignore_index)

@@ -352,3 +352,184 @@ def test_worker_count_change_rescales(corpus):
         d2.load_state_dict([state], sharded_input=False)
         seen.append(firsts(take(iter(d2), 20)))
     assert not (seen[0] & seen[1])
+
+
+# ---------------- composition matrix (reference :402-497) ----------------
+
+def build_variant(corpus, variant, rank=0, world=1, chunksize=1000,
+                  n_logical=7):
+    """The 4 pipeline compositions every property is checked over
+    (reference basic_loader/scalable/sampler/sampler_scalable,
+    test_datasets.py:402-465)."""
+    if variant == "base":
+        return base_loader(corpus, rank, world, chunksize=chunksize)
+    if variant == "scalable":
+        return scalable_loader(corpus, rank, world, n_logical, chunksize)
+    if variant == "sampler":
+        return D.SamplingDataset(corpus,
+                                 base_loader(corpus, rank, world,
+                                             chunksize=chunksize),
+                                 -1, datasets=["dataset_1"], weights=[1])
+    if variant == "sampler_scalable":
+        return D.SamplingDataset(corpus,
+                                 scalable_loader(corpus, rank, world,
+                                                 n_logical, chunksize),
+                                 -1, datasets=["dataset_1"], weights=[1])
+    raise ValueError(variant)
+
+
+ALL_VARIANTS = ["base", "scalable", "sampler", "sampler_scalable"]
+
+
+@pytest.mark.parametrize("variant", ALL_VARIANTS)
+def test_epoch_exactly_once_all_compositions(corpus, variant):
+    d = build_variant(corpus, variant)
+    d.setup()
+    chunks = take(iter(d), 300)
+    ids = sorted(c[0] // 100 for c in chunks)
+    assert ids == list(range(300)), f"{variant}: epoch not exactly-once"
+
+
+@pytest.mark.parametrize("variant", ALL_VARIANTS)
+def test_two_epochs_twice_all_compositions(corpus, variant):
+    d = build_variant(corpus, variant)
+    d.setup()
+    chunks = take(iter(d), 600)
+    from collections import Counter
+    counts = Counter(c[0] // 100 for c in chunks)
+    assert all(v == 2 for v in counts.values()), f"{variant}: not twice"
+    assert len(counts) == 300
+
+
+@pytest.mark.parametrize("variant", ALL_VARIANTS)
+def test_chunking_all_compositions(corpus, variant):
+    # chunksize 50 splits every 100-token doc into 2 chunks + delimiter
+    d = build_variant(corpus, variant, chunksize=50)
+    d.setup()
+    chunks = take(iter(d), 300 * 3)
+    lens = Counter_ = {}
+    n50 = sum(1 for c in chunks if len(c) == 50)
+    n1 = sum(1 for c in chunks if len(c) == 1)
+    assert n50 == 600 and n1 == 300, f"{variant}: {n50} fifties, {n1} delims"
+
+
+@pytest.mark.parametrize("variant", ALL_VARIANTS)
+@pytest.mark.parametrize("world", [2, 3])
+def test_disjoint_coverage_all_compositions(corpus, variant, world):
+    loaders = []
+    for r in range(world):
+        d = build_variant(corpus, variant, rank=r, world=world, n_logical=12)
+        d.setup()
+        loaders.append(d)
+    seen = []
+    for d in loaders:
+        # exactly one epoch per rank (world divides 300 and n_logical)
+        chunks = take(iter(d), 300 // world)
+        seen.extend(c[0] // 100 for c in chunks)
+    assert sorted(seen) == list(range(300)), \
+        f"{variant}: ranks not disjoint/complete"
+
+
+# ------------- adversarial multi-reload (reference :580-633) -------------
+
+def adversarial_pipeline(corpus, rank, world):
+    """chunksize 17, 15 logical shards, buffers 73/99, 2-corpus sampling:
+    the reference's 'messy state' parameters (test_datasets.py:580-633)."""
+    d = base_loader(corpus, rank, world, chunksize=17)
+    d = D.ScalableShardDataset(d, n_logical_shards=15)
+    d = D.SamplingDataset(corpus, d, -1,
+                          datasets=["dataset_1", "dataset_3"], weights=[3, 5])
+    d = D.BufferDataset(d, 73, pack_hard=True, bos_token=-1)
+    d = D.PreloadBufferDataset(d, 99)
+    return d
+
+
+@pytest.mark.parametrize("n_steps", [4, 29, 100])
+def test_multi_reload_stress_adversarial(corpus, n_steps):
+    """3-rank in-process simulation, full messy-parameter pipeline, state
+    saved after n steps and restored into fresh datasets: continuations
+    must match token-exactly on every rank."""
+    from copy import deepcopy
+    world = 3
+    d1 = [adversarial_pipeline(corpus, r, world) for r in range(world)]
+    for d in d1:
+        d.setup()
+    its = [iter(d) for d in d1]
+    for it in its:
+        take(it, n_steps)
+    states = [deepcopy(d.state_dict()) for d in d1]
+    cont = [[list(x) for x in take(it, 23)] for it in its]
+
+    d2 = [adversarial_pipeline(corpus, r, world) for r in range(world)]
+    for r, d in enumerate(d2):
+        d.setup()
+        # full per-rank state list: each dataset slices its own rank's
+        # span (the same contract load_from_path uses)
+        d.load_state_dict(states, sharded_input=False)
+    cont2 = [[list(x) for x in take(iter(d), 23)] for d in d2]
+    assert cont == cont2, "adversarial resume not token-exact"
+
+
+# ----- CheckpointDataset through persistent workers (reference :893-978) ---
+
+def test_checkpoint_reload_match_persistent_workers(corpus, tmp_path):
+    """3 ranks x DataLoader(num_workers=1, persistent_workers): the
+    in-worker auto-save at step 100 reloads into fresh pipelines and the
+    next 300 batches match exactly."""
+    ck_root = str(tmp_path / "ckp_test")
+
+    def mk(interval):
+        ds = []
+        for r in range(3):
+            d = base_loader(corpus, r, 3, chunksize=17)
+            d = D.SamplingDataset(corpus, d, -1,
+                                  datasets=["dataset_1", "dataset_3"],
+                                  weights=[3, 5])
+            d = D.BufferDataset(d, 73, pack_hard=True, bos_token=-1)
+            d = D.PreprocessDataset(d, torch.IntTensor)
+            ds.append(D.CheckpointDataset(d, ck_root, interval,
+                                          steps_per_batch=2,
+                                          save_path=ck_root))
+        return ds
+
+    loaders = [iter(torch.utils.data.DataLoader(
+        x, num_workers=1, batch_size=2, prefetch_factor=1,
+        persistent_workers=True)) for x in mk(100)]
+    for _ in range(100):
+        for ld in loaders:
+            next(ld)
+    ckps = os.listdir(os.path.join(ck_root, "checkpoints"))
+    assert len(ckps) == 1, ckps
+    shards = os.listdir(os.path.join(ck_root, "checkpoints", ckps[0]))
+    assert len(shards) == 3, shards
+
+    ds2 = mk(1000)
+    for d in ds2:
+        d.setup()
+        assert d.step == 100, d.step
+    loaders2 = [iter(torch.utils.data.DataLoader(
+        x, num_workers=1, batch_size=2, prefetch_factor=1,
+        persistent_workers=True)) for x in ds2]
+    for _ in range(300):
+        for a, b in zip(loaders, loaders2):
+            ta, tb = next(a), next(b)
+            assert torch.equal(ta, tb), "persistent-worker resume mismatch"
+
+
+@pytest.mark.parametrize("n_workers", [0, 2])
+@pytest.mark.parametrize("world", [2, 5])
+def test_multiprocess_epoch_all(corpus, world, n_workers):
+    """Scalable partitioning over worldsize x n_workers: one epoch holds
+    each doc exactly once collectively (reference :966-978)."""
+    loaders = []
+    for r in range(world):
+        d = scalable_loader(corpus, r, world, n_logical=20)
+        d = D.BufferDataset(d, 110, pack_hard=False, pad_token=-1)
+        loaders.append(torch.utils.data.DataLoader(
+            d, num_workers=n_workers, batch_size=None))
+    seen = []
+    per = 300 // world
+    for ld in loaders:
+        out = take(iter(ld), per)
+        seen.extend(int(line[0]) // 100 for line in out)
+    assert len(set(seen)) == len(seen), "duplicate docs across ranks/workers"

@@ -446,3 +446,177 @@ def test_checkpointer_resume_real_pipeline(tmp_path):
     for (ex, ey), (gx, gy) in zip(expect, got):
         assert torch.equal(ex, gx) and torch.equal(ey, gy), \
             "resume not token-exact"
+
+
+def _worker70(rank, world, port, strat, intra, q, steps=3):
+    """70B-execution-path proxy: reshard_after_forward=True (transient
+    flat params AND grads) + selective AC + streamed meta init, at the
+    8-rank topology of the driver's single-node scaling run."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from fms_fsdp_amd.models import Llama, LlamaBlock
+        from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+        from fms_fsdp_amd.parallel.policies import apply_selective_ac
+        with torch.device("meta"):
+            m = Llama(_small_cfg())
+        apply_selective_ac(m, LlamaBlock, "1/2")
+        torch.manual_seed(0)
+        sm = ShardedModel(m, LlamaBlock, sharding_strategy=strat,
+                          param_dtype=torch.float32,
+                          reshard_after_forward=True,
+                          intra_node_size=intra)
+        opt = ShardedAdamW(sm, lr=1e-3)
+        g = torch.Generator().manual_seed(42)
+        x = torch.randint(0, 128, (2, 32), generator=g)
+        y = torch.randint(0, 128, (2, 32), generator=g)
+        losses = []
+        for _ in range(steps):
+            opt.zero_grad()
+            loss = sm(x, labels=y)
+            loss.backward()
+            sm.clip_grad_norm_(1.0)
+            opt.step()
+            losses.append(loss.item())
+        # transient buffers really were freed after backward
+        for u in sm.units:
+            assert u.flat_param.untyped_storage().size() == 0, u.name
+            assert u.flat_grad.untyped_storage().size() == 0, u.name
+        if rank == 0:
+            q.put(("ok", losses))
+    except Exception as e:
+        q.put(("err", f"{type(e).__name__}: {e}"))
+        raise
+    finally:
+        dist.destroy_process_group()
+
+
+def _ref70():
+    """Single-process trajectory with the same init path (meta+seed)."""
+    from fms_fsdp_amd.models import Llama, LlamaBlock
+    from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+    with torch.device("meta"):
+        m = Llama(_small_cfg())
+    torch.manual_seed(0)
+    sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                      param_dtype=torch.float32)
+    opt = ShardedAdamW(sm, lr=1e-3)
+    g = torch.Generator().manual_seed(42)
+    x = torch.randint(0, 128, (2, 32), generator=g)
+    y = torch.randint(0, 128, (2, 32), generator=g)
+    losses = []
+    for _ in range(3):
+        opt.zero_grad()
+        loss = sm(x, labels=y)
+        loss.backward()
+        sm.clip_grad_norm_(1.0)
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
+@pytest.mark.parametrize("strat,intra,port", [
+    ("fsdp", None, PORT + 41),
+    ("hsdp", 4, PORT + 42),   # 2 'nodes' x 4 'GPUs'
+])
+def test_world8_70b_execution_path(strat, intra, port):
+    """World-8 gloo: meta-streamed init + reshard_after_forward +
+    selective AC + (fsdp | hsdp 2x4) must match the single-process
+    trajectory — the code-level readiness check for the 70B row
+    (VERDICT weak #6)."""
+    ref = _ref70()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_worker70, args=(8, port, strat, intra, q), nprocs=8, join=True)
+    status, got = q.get()
+    assert status == "ok", got
+    for a, b in zip(ref, got):
+        assert abs(a - b) < 1e-5, (ref, got)
+
+
+def _ckpt8_worker(rank, world, port, tmpdir, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from fms_fsdp_amd.models import Llama, LlamaBlock
+        from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+        from fms_fsdp_amd.utils.checkpointing import Checkpointer
+        torch.manual_seed(0)
+        m = Llama(_small_cfg())
+        m.reset_parameters()
+        sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                          param_dtype=torch.float32,
+                          reshard_after_forward=True)
+        opt = ShardedAdamW(sm, lr=1e-3)
+        g = torch.Generator().manual_seed(42)
+        x = torch.randint(0, 128, (2, 32), generator=g)
+        y = torch.randint(0, 128, (2, 32), generator=g)
+        for _ in range(2):
+            opt.zero_grad()
+            sm(x, labels=y).backward()
+            sm.clip_grad_norm_(1.0)
+            opt.step()
+        Checkpointer(tmpdir, 3, "fsdp", rank, rank).save(2, sm, opt, None)
+        if rank == 0:
+            q.put(("ok", None))
+    except Exception as e:
+        q.put(("err", f"{type(e).__name__}: {e}"))
+        raise
+    finally:
+        dist.destroy_process_group()
+
+
+def _resume2_worker(rank, world, port, tmpdir, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from fms_fsdp_amd.models import Llama, LlamaBlock
+        from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+        from fms_fsdp_amd.utils.checkpointing import Checkpointer
+        torch.manual_seed(77)   # different init: must be overwritten
+        m = Llama(_small_cfg())
+        m.reset_parameters()
+        sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                          param_dtype=torch.float32)
+        opt = ShardedAdamW(sm, lr=1e-3)
+        ck = Checkpointer(tmpdir, 3, "fsdp", rank, rank)
+        _, _, _, step, _, resuming = ck.load(sm, opt, None, path="")
+        assert step == 2 and resuming
+        g = torch.Generator().manual_seed(42)
+        x = torch.randint(0, 128, (2, 32), generator=g)
+        y = torch.randint(0, 128, (2, 32), generator=g)
+        opt.zero_grad()
+        loss = sm(x, labels=y)
+        loss.backward()
+        sm.clip_grad_norm_(1.0)
+        opt.step()
+        if rank == 0:
+            q.put(("ok", loss.item()))
+    except Exception as e:
+        q.put(("err", f"{type(e).__name__}: {e}"))
+        raise
+    finally:
+        dist.destroy_process_group()
+
+
+def test_world8_save_reshard2_resume(tmp_path):
+    """Save at world=8, resume at world=2 (8->2 shard resharding incl.
+    optimizer moments), continue a step: loss must equal the
+    uninterrupted single-process step-3 loss."""
+    ref3 = _single_process_reference()[2]
+    tmpdir = str(tmp_path)
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_ckpt8_worker, args=(8, PORT + 43, tmpdir, q), nprocs=8,
+             join=True)
+    status, _ = q.get()
+    assert status == "ok", _
+    q2 = ctx.SimpleQueue()
+    mp.spawn(_resume2_worker, args=(2, PORT + 44, tmpdir, q2), nprocs=2,
+             join=True)
+    status, loss3 = q2.get()
+    assert status == "ok", loss3
+    assert abs(loss3 - ref3) < 1e-5, (loss3, ref3)
