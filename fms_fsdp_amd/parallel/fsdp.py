@@ -165,8 +165,8 @@ class FlatUnit:
         grads alone are 140 GB if kept resident — they must be transient).
         The buffer is re-allocated and zeroed in prepare_grads() at the
         unit's next backward-pre."""
-        if self.grad_shard is self.flat_grad:
-            return
+        if self.S == 1 or self.grad_shard is self.flat_grad:
+            return  # S==1: flat_grad may be a pooled view — never resize
         if rs_stream is not None:
             # the caching allocator must not hand this block out until
             # the reduce-scatter reading it on rs_stream has passed
@@ -358,12 +358,92 @@ class ShardedModel(nn.Module):
             # materialize shards (broadcast rank-0 init for determinism)
             for u in self.all_units:
                 u.materialize(src_rank_broadcast=True)
+        self._pool_unit_storage()
 
         # streams
         self.comm_stream = torch.cuda.Stream() if self.is_cuda else None
         self.rs_stream = torch.cuda.Stream() if self.is_cuda else None
 
         self._install_hooks()
+
+    def _pool_unit_storage(self):
+        """Repack per-unit persistent tensors into single contiguous pools
+        (units keep views). The optimizer phase then runs as ONE fused
+        AdamW launch + ONE sq-norm launch over the whole model instead of
+        2 launches x n_units (the reference pays a foreach launch per
+        param group; at 7B/33 units the per-unit scheme measured ~66
+        launches per step in profiles/7b_1gpu_step_r01_final.md).
+
+        Pooled in every regime: master_shard, exp_avg, exp_avg_sq (always
+        persistent shard-size buffers). S>1 additionally pools
+        param_shard / grad_shard (persistent even under
+        reshard_after_forward — only the full flat buffers are
+        transient). S==1 pools the flat param/grad buffers themselves
+        (they are never freed at S==1).
+        """
+        units = self.all_units
+        if not units:
+            return
+        dev = self.device
+        S = units[0].S
+
+        def pool(attr, dtype, sizes):
+            total = sum(sizes)
+            buf = torch.empty(total, dtype=dtype, device=dev)
+            off = 0
+            views = []
+            for u, n in zip(units, sizes):
+                v = buf[off:off + n]
+                v.copy_(getattr(u, attr))
+                views.append(v)
+                off += n
+            return buf, views
+
+        sizes = [u.shard_size for u in units]
+        for attr, dt in (("master_shard", torch.float32),
+                         ("exp_avg", torch.float32),
+                         ("exp_avg_sq", torch.float32)):
+            buf, views = pool(attr, dt, sizes)
+            setattr(self, "_pool_" + attr, buf)
+            for u, v in zip(units, views):
+                setattr(u, attr, v)
+        if S > 1:
+            pdt = units[0].param_dtype
+            rdt = units[0].reduce_dtype
+            self._pool_param_shard, views = pool("param_shard", pdt, sizes)
+            for u, v in zip(units, views):
+                u.param_shard = v
+            self._pool_grad_shard, views = pool("grad_shard", rdt, sizes)
+            for u, v in zip(units, views):
+                u.grad_shard = v
+        else:
+            totals = [u.total for u in units]
+            pdt = units[0].param_dtype
+            self._pool_param_shard, views = pool("flat_param", pdt, totals)
+            for u, v in zip(units, views):
+                u.flat_param = v
+                u.param_shard = v
+                u._point_params_into_flat()
+            self._pool_grad_shard, gviews = pool("flat_grad", pdt, totals)
+            rdt = units[0].reduce_dtype
+            sep_reduce = rdt != pdt
+            for u, v in zip(units, gviews):
+                u.flat_grad = v
+                if not sep_reduce:
+                    u.grad_shard = v   # keep the S==1 alias invariant
+                u._set_grad_views()
+            if sep_reduce:
+                self._pool_reduce_shard, views = pool("grad_shard", rdt, totals)
+                for u, v in zip(units, views):
+                    u.grad_shard = v
+
+    def _pooled_grad_norm_src(self):
+        """The single tensor clip_grad_norm_ sums squares over."""
+        units = self.all_units
+        if units and units[0].S == 1 and \
+                units[0].grad_shard is not units[0].flat_grad:
+            return self._pool_reduce_shard
+        return self._pool_grad_shard
 
     def _materialize_from_meta(self):
         """Materialize units one at a time from a meta-device model.
@@ -507,13 +587,14 @@ class ShardedModel(nn.Module):
     # ---------------- training utilities ----------------
 
     def clip_grad_norm_(self, max_norm):
-        """Global grad-norm over the bf16 grad shards (reference:
+        """Global grad-norm over the grad shards (reference:
         train_utils.py:96 model.clip_grad_norm_). The clip factor is NOT
         applied as an extra pass over the grads — it is folded into the
-        fused AdamW kernel (one less 2x-shard-size memory sweep)."""
+        fused AdamW kernel (one less 2x-shard-size memory sweep). Thanks
+        to the pooled shard storage this is ONE kernel launch."""
         for u in self.all_units:
             u.wait_grads()
-        local = ops.sq_norm([u.grad_shard for u in self.all_units])
+        local = ops.sq_norm([self._pooled_grad_norm_src()])
         if self.shard_group is not None:
             dist.all_reduce(local, group=self.shard_group)
         total_norm = local.sqrt()
@@ -572,16 +653,22 @@ class ShardedAdamW:
         lr = self.param_groups[0]["lr"]
         b1, b2 = self.betas
         gscale = getattr(self.m, "_clip_coef", None)
-        for u in self.m.all_units:
+        m = self.m
+        for u in m.all_units:
             u.wait_grads()
-            lowp = u.param_shard.dtype in (torch.bfloat16, torch.float16)
-            published = ops.fused_adamw(
-                u.master_shard, u.grad_shard, u.exp_avg, u.exp_avg_sq,
-                self.step_count, lr, b1, b2, self.eps, self.weight_decay,
-                grad_scale=gscale,
-                p_bf16_out=u.param_shard if lowp else None)
-            if not published or not lowp:
-                u.publish_master_to_shard()
+        # pooled storage: the whole model updates in ONE fused launch
+        pdt = m._pool_param_shard.dtype
+        lowp = pdt in (torch.bfloat16, torch.float16)
+        published = ops.fused_adamw(
+            m._pool_master_shard, m._pooled_grad_norm_src(),
+            m._pool_exp_avg, m._pool_exp_avg_sq,
+            self.step_count, lr, b1, b2, self.eps, self.weight_decay,
+            grad_scale=gscale,
+            p_bf16_out=m._pool_param_shard if lowp else None)
+        if not published or not lowp:
+            with torch.no_grad():
+                m._pool_param_shard.copy_(m._pool_master_shard)
+        for u in m.all_units:
             u.mark_stale()
 
     def zero_grad(self, set_to_none=False):
