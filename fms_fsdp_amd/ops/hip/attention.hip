@@ -119,13 +119,54 @@ __device__ __forceinline__ TR4 tr_read4(const char* base, int off0, int off1,
   return out;
 }
 
+// ---- immediate-offset, software-pipelined transpose reads ------------
+// The 16 tr-read addresses of a PV/dq/dkv output loop share ONE runtime
+// base (swizzle ^ half-select ^ 16-lane-block, all lane-derived); the
+// remaining terms are compile-time. Issuing through `offset:` immediates
+// removes the per-read VALU address chain (PMC round-1: VALU is the
+// limiter, profiles/attn_pmc_counters_r01.md), and splitting issue from
+// wait lets fragment t+1 stream from LDS while fragment t feeds mfmas.
+// Counted waits are safe here because DS ops retire IN ORDER: "allow 4
+// outstanding" implies everything older (incl. staging ds_writes) is
+// done. No SMEM is issued inside these windows.
+template <int O0, int O1, int O2, int O3>
+__device__ __forceinline__ void tr_issue4(unsigned base, U2x64& r0,
+                                          U2x64& r1) {
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4 offset:%c5\n\t"
+      "ds_read_b64_tr_b16 %1, %4 offset:%c6\n\t"
+      "ds_read_b64_tr_b16 %2, %4 offset:%c7\n\t"
+      "ds_read_b64_tr_b16 %3, %4 offset:%c8"
+      : "=&v"(r0.u[0]), "=&v"(r0.u[1]), "=&v"(r1.u[0]), "=&v"(r1.u[1])
+      : "v"(base), "i"(O0), "i"(O1), "i"(O2), "i"(O3)
+      : "memory");
+}
+
+template <int CNT>
+__device__ __forceinline__ void tr_wait(U2x64& r0, U2x64& r1) {
+  asm volatile("s_waitcnt lgkmcnt(%c4)"
+               : "+v"(r0.u[0]), "+v"(r0.u[1]), "+v"(r1.u[0]), "+v"(r1.u[1])
+               : "i"(CNT)
+               : "memory");
+}
+
+// lane-derived base for the subtiled-image transpose reads: row-group
+// XOR (constant 16*hb for every fragment of these loops), the in-subtile
+// lane column, the hb*8 key offset (rg = hb*2 -> 256*hb bytes) and the
+// 16-lane block select. R4B = R4*128 bytes per column block.
+template <int R4B>
+__device__ __forceinline__ unsigned tr_base(int lane, int col, int hb) {
+  return (unsigned)((((lane & 15) * 8) ^ (16 * hb)) + hb * 256 +
+                    (col >> 4) * R4B);
+}
+
 
 // ---------------------------------------------------------------------
 // Forward. Block = 4 waves x 32 q-rows = 128 q rows; KV tile = 64 keys.
 // grid.x = s/128, grid.y = b*h.
 // ---------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
+__global__ __launch_bounds__(256, 3) void attn_fwd_kernel(
     const short* __restrict__ qg, const short* __restrict__ kg,
     const short* __restrict__ vg, short* __restrict__ og,
     float* __restrict__ lseg, int B, int S, int H, int KVH, float scale,
@@ -134,12 +175,16 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   constexpr int NC = D / 16;   // QK^T k-chunks
   constexpr int NT = D / 32;   // 32-wide output tiles
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // double-buffered SUBTILED K/V images (same layout as the backward
-  // kernels: coalesced b128 staging, plain reads for the S^T A-operand,
-  // hardware transpose reads for the PV A-operand). ONE barrier/tile.
+  // SUBTILED K/V images (same layout as the backward kernels: coalesced
+  // b128 staging, plain reads for the S^T A-operand, hardware transpose
+  // reads for the PV A-operand). LDS diet for occupancy: K is
+  // SINGLE-buffered, V double-buffered (48 KB at D=128 -> 3 blocks/CU,
+  // up from 2 at the fully double-buffered 64 KB; VGPRs fit 3 waves).
+  // K(t+1) is staged mid-tile right after the last S-phase read of
+  // K(t), costing a second barrier per tile.
   constexpr int KB = KVB * D * 2;
-#define KLDS(buf) (smem + ((buf) ? 2 * KB : 0))
-#define VLDS(buf) (smem + KB + ((buf) ? 2 * KB : 0))
+#define KLDS() (smem)
+#define VLDS(buf) (smem + KB + ((buf) ? KB : 0))
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -181,47 +226,70 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
   const int t256 = threadIdx.x;
   // subtiled staging: thread t: key = t&63, colblk pair = t>>6 (each
-  // thread copies 2x16B per image; 64-key rows x D cols)
-  auto stage = [&](int buf, int kv0) {
-    const int key = t256 & 63;
-    const int d0 = (t256 >> 6) * (D / 4);   // 32 cols per thread @D=128
+  // thread copies 2x16B per image; 64-key rows x D cols). Global
+  // pointers advance incrementally (no per-tile 64-bit muls) and the
+  // LDS offsets are per-thread constants computed once.
+  const int skey = t256 & 63;
+  const int sd0 = (t256 >> 6) * (D / 4);   // 32 cols per thread @D=128
+  const short* kp_s = kbase + (long long)skey * krow_stride + sd0;
+  const short* vp_s = vbase + (long long)skey * vstride + sd0;
+  const long long kstep = (long long)KVB * krow_stride;
+  const long long vstep = (long long)KVB * vstride;
+  int soff[D / 64][2];
+#pragma unroll
+  for (int c2 = 0; c2 < D / 64; ++c2) {
+    soff[c2][0] = SUBT_OFF(skey, sd0 + c2 * 16, 16);
+    soff[c2][1] = SUBT_OFF(skey, sd0 + c2 * 16 + 8, 16);
+  }
+  auto stage_k = [&]() {
 #pragma unroll
     for (int c2 = 0; c2 < D / 64; ++c2) {
-      const int dc = d0 + c2 * 16;
-      const long long gk = (long long)(kv0 + key) * krow_stride + dc;
-      const long long gv = (long long)(kv0 + key) * vstride + dc;
-      const int off0 = SUBT_OFF(key, dc, 16);
-      const int off1 = SUBT_OFF(key, dc + 8, 16);
-      *(f32x4*)(KLDS(buf) + off0) = *(const f32x4*)(kbase + gk);
-      *(f32x4*)(KLDS(buf) + off1) = *(const f32x4*)(kbase + gk + 8);
-      *(f32x4*)(VLDS(buf) + off0) = *(const f32x4*)(vbase + gv);
-      *(f32x4*)(VLDS(buf) + off1) = *(const f32x4*)(vbase + gv + 8);
+      *(f32x4*)(KLDS() + soff[c2][0]) = *(const f32x4*)(kp_s + c2 * 16);
+      *(f32x4*)(KLDS() + soff[c2][1]) = *(const f32x4*)(kp_s + c2 * 16 + 8);
     }
+    kp_s += kstep;
   };
-  stage(0, 0);
+  auto stage_v = [&](int buf) {
+#pragma unroll
+    for (int c2 = 0; c2 < D / 64; ++c2) {
+      *(f32x4*)(VLDS(buf) + soff[c2][0]) = *(const f32x4*)(vp_s + c2 * 16);
+      *(f32x4*)(VLDS(buf) + soff[c2][1]) = *(const f32x4*)(vp_s + c2 * 16 + 8);
+    }
+    vp_s += vstep;
+  };
+  stage_k();
+  stage_v(0);
   __syncthreads();
+  // per-buffer tr-read bases for the PV loops (R4 = KVB/4 = 16)
+  const unsigned pv_swz = tr_base<2048>(lane, col, hb);
+  const unsigned pvb[2] = {(unsigned)(unsigned long long)VLDS(0) + pv_swz,
+                           (unsigned)(unsigned long long)VLDS(1) + pv_swz};
   int cur = 0;
   for (int tile = 0; tile < ntiles; ++tile) {
     const int kv0 = tile * KVB;
-    if (tile + 1 < ntiles) stage(cur ^ 1, (tile + 1) * KVB);
+    if (tile + 1 < ntiles) stage_v(cur ^ 1);
 
     // ---- two 32-key sub-tiles, each with its own online-softmax pass.
-    // Register economy: one live accS/p set (16 regs) instead of two,
-    // keeping total VGPR+AGPR under 256 for 2 waves/SIMD occupancy.
-    // defer-max (guide T13, THR=8 nats = 11.5 bits): the O/l rescale runs only when the
-    // sub-tile max exceeds the running max by more than THR; P is then
-    // bounded by e^THR which the fp32 accumulate tolerates. Decision is
-    // made BEFORE this sub-tile's P is exponentiated (the safe order).
+    // Register economy: one live accS/p set (16 regs) instead of two.
+    // defer-max (guide T13, THR=8 nats = 11.5 bits): the O/l rescale
+    // runs only when the sub-tile max exceeds the running max by more
+    // than THR; P is then bounded by e^THR which the fp32 accumulate
+    // tolerates. Decision is made BEFORE this sub-tile's P is
+    // exponentiated (the safe order).
+    // S-phase + softmax + pack for one sub-tile; PV split out so kt=1's
+    // PV can run after the mid-tile K barrier (single-buffered K).
+    bf16x8v pb1_0, pb1_1;
+    const bool act1 = (kv0 + 32) <= qw + 31;   // wave-uniform causal
 #pragma unroll
     for (int kt = 0; kt < 2; ++kt) {
       const int kv32 = kv0 + kt * 32;
-      if (kv32 > qw + 31) break;      // wave-uniform causal tile skip
+      if (kt == 1 && !act1) break;
       f32x16 accS = (f32x16)(0.f);
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         const int row = kt * 32 + col;
         const bf16x8v a = *(const bf16x8v*)(
-            KLDS(cur) + SUBT_OFF(row, c * 16 + hb * 8, 16));
+            KLDS() + SUBT_OFF(row, c * 16 + hb * 8, 16));
         accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qb[c], accS, 0, 0, 0);
       }
       float p[16];
@@ -252,31 +320,72 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       }
       l_run = l_run * alpha + s_own + __shfl_xor(s_own, 32, 64);
 
-      // PV for this sub-tile: A = V^T via hardware transpose reads on
-      // the row-major V image; keys kv32..kv32+31 = chunks 2kt, 2kt+1
-      const bf16x8v pb0 = pack_pT_chunk(p);
-      const bf16x8v pb1 = pack_pT_chunk(p + 8);
+      const unsigned pvbase = pvb[cur] + kt * 1024;   // rg += kt*8
+      if (kt == 0) {
+        // PV immediately (software-pipelined immediate-offset reads;
+        // first issue BEFORE the P pack so LDS latency hides behind
+        // the pack VALU work)
+        U2x64 fr[2][2];
+        tr_issue4<0, 128, 512, 640>(pvbase, fr[0][0], fr[0][1]);
+        const bf16x8v pb0 = pack_pT_chunk(p);
+        const bf16x8v pb1 = pack_pT_chunk(p + 8);
+#pragma unroll
+        for (int t = 0; t < NT; ++t) {
+          const int pp = t & 1;
+          if (t + 1 < NT) {
+            // offsets: (t+1)*2*(KVB/4)*128 = (t+1)*4096 + rg*128
+            if (t + 1 == 1)
+              tr_issue4<4096, 4096 + 128, 4096 + 512, 4096 + 640>(
+                  pvbase, fr[1][0], fr[1][1]);
+            else if (t + 1 == 2)
+              tr_issue4<8192, 8192 + 128, 8192 + 512, 8192 + 640>(
+                  pvbase, fr[0][0], fr[0][1]);
+            else
+              tr_issue4<12288, 12288 + 128, 12288 + 512, 12288 + 640>(
+                  pvbase, fr[1][0], fr[1][1]);
+            tr_wait<4>(fr[pp][0], fr[pp][1]);
+          } else {
+            tr_wait<0>(fr[pp][0], fr[pp][1]);
+          }
+          accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][0].v, pb0,
+                                                            accO[t], 0, 0, 0);
+          accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][1].v, pb1,
+                                                            accO[t], 0, 0, 0);
+        }
+      } else {
+        // pack only; PV runs after the K barrier below
+        pb1_0 = pack_pT_chunk(p);
+        pb1_1 = pack_pT_chunk(p + 8);
+      }
+    }
+    // all waves are done reading K(tile): restage the single K image
+    __syncthreads();
+    if (tile + 1 < ntiles) stage_k();
+    if (act1) {
+      const unsigned pvbase = pvb[cur] + 1024;
+      U2x64 fr[2][2];
+      tr_issue4<0, 128, 512, 640>(pvbase, fr[0][0], fr[0][1]);
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
-        const int dvblk = t * 2 + (col >> 4);
-        const int ka = kt * 32 + hb * 8;        // kc2 = 0 chunk
-        const int kb = kt * 32 + 16 + hb * 8;   // kc2 = 1 chunk
-        const int rga0 = ka >> 2, rga1 = (ka + 4) >> 2;
-        const int rgb0 = kb >> 2, rgb1 = (kb + 4) >> 2;
-        const TR4 f = tr_read4(
-            VLDS(cur),
-            (dvblk * (KVB / 4) + rga0) * 128 +
-                (((lane & 15) * 8) ^ ((rga0 & 2) << 3)),
-            (dvblk * (KVB / 4) + rga1) * 128 +
-                (((lane & 15) * 8) ^ ((rga1 & 2) << 3)),
-            (dvblk * (KVB / 4) + rgb0) * 128 +
-                (((lane & 15) * 8) ^ ((rgb0 & 2) << 3)),
-            (dvblk * (KVB / 4) + rgb1) * 128 +
-                (((lane & 15) * 8) ^ ((rgb1 & 2) << 3)));
-        accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(f.a, pb0, accO[t],
-                                                          0, 0, 0);
-        accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(f.b, pb1, accO[t],
-                                                          0, 0, 0);
+        const int pp = t & 1;
+        if (t + 1 < NT) {
+          if (t + 1 == 1)
+            tr_issue4<4096, 4096 + 128, 4096 + 512, 4096 + 640>(
+                pvbase, fr[1][0], fr[1][1]);
+          else if (t + 1 == 2)
+            tr_issue4<8192, 8192 + 128, 8192 + 512, 8192 + 640>(
+                pvbase, fr[0][0], fr[0][1]);
+          else
+            tr_issue4<12288, 12288 + 128, 12288 + 512, 12288 + 640>(
+                pvbase, fr[1][0], fr[1][1]);
+          tr_wait<4>(fr[pp][0], fr[pp][1]);
+        } else {
+          tr_wait<0>(fr[pp][0], fr[pp][1]);
+        }
+        accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][0].v, pb1_0,
+                                                          accO[t], 0, 0, 0);
+        accO[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][1].v, pb1_1,
+                                                          accO[t], 0, 0, 0);
       }
     }
     __syncthreads();
@@ -401,33 +510,38 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   const int t256 = threadIdx.x;
   // stage K/V into subtiled images: thread t owns key t&31 and ADJACENT
   // column chunks (one cacheline per image per thread — a generic
-  // strided-chunk loop here cost dq +16%/dkv +40%, profiles/)
-  auto stage = [&](int buf, int kv0s) {
-    const int key = t256 & 31;
+  // strided-chunk loop here cost dq +16%/dkv +40%, profiles/).
+  // Incremental global pointers + precomputed LDS offsets (see fwd).
+  const int skey = t256 & 31;
+  const int sd0 = (t256 >> 5) * ((D == 128) ? 16 : 8);
+  const short* kp_s = kbase + (long long)skey * krow_stride + sd0;
+  const short* vp_s = vbase + (long long)skey * vstride + sd0;
+  const long long kstep = (long long)KVB * krow_stride;
+  const long long vstep = (long long)KVB * vstride;
+  const int soff0 = SUBT_OFF(skey, sd0, 8);
+  const int soff1 = SUBT_OFF(skey, sd0 + 8, 8);
+  auto stage = [&](int buf) {
     if constexpr (D == 128) {
-      const int d0 = (t256 >> 5) * 16;
-      const long long gk = (long long)(kv0s + key) * krow_stride + d0;
-      const long long gv = (long long)(kv0s + key) * vstride + d0;
-      const int off0 = SUBT_OFF(key, d0, 8);
-      const int off1 = SUBT_OFF(key, d0 + 8, 8);
-      *(f32x4*)(KIMG(buf) + off0) = *(const f32x4*)(kbase + gk);
-      *(f32x4*)(KIMG(buf) + off1) = *(const f32x4*)(kbase + gk + 8);
-      *(f32x4*)(VIMG(buf) + off0) = *(const f32x4*)(vbase + gv);
-      *(f32x4*)(VIMG(buf) + off1) = *(const f32x4*)(vbase + gv + 8);
+      *(f32x4*)(KIMG(buf) + soff0) = *(const f32x4*)(kp_s);
+      *(f32x4*)(KIMG(buf) + soff1) = *(const f32x4*)(kp_s + 8);
+      *(f32x4*)(VIMG(buf) + soff0) = *(const f32x4*)(vp_s);
+      *(f32x4*)(VIMG(buf) + soff1) = *(const f32x4*)(vp_s + 8);
     } else {  // D=64: 32 keys x 8 chunks, exactly one 16B chunk/thread
-      const int d0 = (t256 >> 5) * 8;
-      const long long gk = (long long)(kv0s + key) * krow_stride + d0;
-      const long long gv = (long long)(kv0s + key) * vstride + d0;
-      const int off = SUBT_OFF(key, d0, 8);
-      *(f32x4*)(KIMG(buf) + off) = *(const f32x4*)(kbase + gk);
-      *(f32x4*)(VIMG(buf) + off) = *(const f32x4*)(vbase + gv);
+      *(f32x4*)(KIMG(buf) + soff0) = *(const f32x4*)(kp_s);
+      *(f32x4*)(VIMG(buf) + soff0) = *(const f32x4*)(vp_s);
     }
+    kp_s += kstep;
+    vp_s += vstep;
   };
-  stage(0, 0);
+  stage(0);
   __syncthreads();
+  // tr-read bases into the K images (R4 = KVB/4 = 8)
+  const unsigned dq_swz = tr_base<1024>(lane, col, hb);
+  const unsigned kib[2] = {(unsigned)(unsigned long long)KIMG(0) + dq_swz,
+                           (unsigned)(unsigned long long)KIMG(1) + dq_swz};
   int cur = 0;
   for (int tile = 0; tile < ntiles; ++tile) {
-    if (tile + 1 < ntiles) stage(cur ^ 1, (tile + 1) * KVB);
+    if (tile + 1 < ntiles) stage(cur ^ 1);
     const int kv0 = tile * KVB;
 
     // S^T and dP^T (A rows = keys, plain subtiled reads)
@@ -451,25 +565,35 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       ds[r] = v;
     }
 
-    // dq^T += K^T @ dS^T : A = tr_read on k image (col dk across key rows)
+    // dq^T += K^T @ dS^T : A = tr_read on k image (col dk across key
+    // rows); pipelined immediate-offset reads (see fwd PV loop)
+    const unsigned kbase_t = kib[cur];
+    U2x64 fr[2][2];
+    tr_issue4<0, 128, 512, 640>(kbase_t, fr[0][0], fr[0][1]);
     bf16x8v dsb[2];
     dsb[0] = pack_pT_chunk(ds);
     dsb[1] = pack_pT_chunk(ds + 8);
 #pragma unroll
     for (int t = 0; t < NT; ++t) {
-      const int dkblk = t * 2 + (col >> 4);
-      const int ka = hb * 8, kb = 16 + hb * 8;
-      const int rga0 = ka >> 2, rga1 = (ka + 4) >> 2;
-      const int rgb0 = kb >> 2, rgb1 = (kb + 4) >> 2;
-      const TR4 f = tr_read4(
-          KIMG(cur),
-          (dkblk * 8 + rga0) * 128 + (((lane & 15) * 8) ^ ((rga0 & 2) << 3)),
-          (dkblk * 8 + rga1) * 128 + (((lane & 15) * 8) ^ ((rga1 & 2) << 3)),
-          (dkblk * 8 + rgb0) * 128 + (((lane & 15) * 8) ^ ((rgb0 & 2) << 3)),
-          (dkblk * 8 + rgb1) * 128 + (((lane & 15) * 8) ^ ((rgb1 & 2) << 3)));
-      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(f.a, dsb[0],
+      const int pp = t & 1;
+      if (t + 1 < NT) {
+        // offsets: (t+1)*2*8*128 = (t+1)*2048 + rg*128
+        if (t + 1 == 1)
+          tr_issue4<2048, 2048 + 128, 2048 + 512, 2048 + 640>(
+              kbase_t, fr[1][0], fr[1][1]);
+        else if (t + 1 == 2)
+          tr_issue4<4096, 4096 + 128, 4096 + 512, 4096 + 640>(
+              kbase_t, fr[0][0], fr[0][1]);
+        else
+          tr_issue4<6144, 6144 + 128, 6144 + 512, 6144 + 640>(
+              kbase_t, fr[1][0], fr[1][1]);
+        tr_wait<4>(fr[pp][0], fr[pp][1]);
+      } else {
+        tr_wait<0>(fr[pp][0], fr[pp][1]);
+      }
+      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][0].v, dsb[0],
                                                          accDQ[t], 0, 0, 0);
-      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(f.b, dsb[1],
+      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][1].v, dsb[1],
                                                          accDQ[t], 0, 0, 0);
     }
     __syncthreads();
@@ -577,32 +701,36 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 
   const int q_start = (blockIdx.x * 128) / 32 * 32;
   const int t256 = threadIdx.x;
-  auto stage = [&](int buf, int q0s) {
-    // thread t owns q row t&31 and ADJACENT column chunks (see the dq
-    // kernel's staging comment: cacheline locality matters here)
-    const int q = t256 & 31;
+  // thread t owns q row t&31 and ADJACENT column chunks (see the dq
+  // kernel's staging comment: cacheline locality matters here);
+  // incremental global pointers + precomputed LDS offsets
+  const int sq = t256 & 31;
+  const int sd0 = (t256 >> 5) * ((D == 128) ? 16 : 8);
+  const short* qp_s = qbase + ((long long)(q_start + sq)) * qrow_stride + sd0;
+  const short* dop_s = dobase + ((long long)(q_start + sq)) * qrow_stride + sd0;
+  const long long qstep = 32 * qrow_stride;
+  const int soff0 = SUBT_OFF(sq, sd0, 8);
+  const int soff1 = SUBT_OFF(sq, sd0 + 8, 8);
+  auto stage = [&](int buf) {
     if constexpr (D == 128) {
-      const int d0 = (t256 >> 5) * 16;
-      const long long g = (long long)(q0s + q) * qrow_stride + d0;
-      const int off0 = SUBT_OFF(q, d0, 8);
-      const int off1 = SUBT_OFF(q, d0 + 8, 8);
-      *(f32x4*)(QIMG(buf) + off0) = *(const f32x4*)(qbase + g);
-      *(f32x4*)(QIMG(buf) + off1) = *(const f32x4*)(qbase + g + 8);
-      *(f32x4*)(DOIMG(buf) + off0) = *(const f32x4*)(dobase + g);
-      *(f32x4*)(DOIMG(buf) + off1) = *(const f32x4*)(dobase + g + 8);
+      *(f32x4*)(QIMG(buf) + soff0) = *(const f32x4*)(qp_s);
+      *(f32x4*)(QIMG(buf) + soff1) = *(const f32x4*)(qp_s + 8);
+      *(f32x4*)(DOIMG(buf) + soff0) = *(const f32x4*)(dop_s);
+      *(f32x4*)(DOIMG(buf) + soff1) = *(const f32x4*)(dop_s + 8);
     } else {  // D=64: 32 rows x 8 chunks, one 16B chunk/thread
-      const int d0 = (t256 >> 5) * 8;
-      const long long g = (long long)(q0s + q) * qrow_stride + d0;
-      const int off = SUBT_OFF(q, d0, 8);
-      *(f32x4*)(QIMG(buf) + off) = *(const f32x4*)(qbase + g);
-      *(f32x4*)(DOIMG(buf) + off) = *(const f32x4*)(dobase + g);
+      *(f32x4*)(QIMG(buf) + soff0) = *(const f32x4*)(qp_s);
+      *(f32x4*)(DOIMG(buf) + soff0) = *(const f32x4*)(dop_s);
     }
+    qp_s += qstep;
+    dop_s += qstep;
   };
-  stage(0, q_start);
+  stage(0);
   __syncthreads();
+  // tr-read base offset into the Q/dO images (R4 = 8)
+  const unsigned kv_swz = tr_base<1024>(lane, col, hb);
   int cur = 0;
   for (int q0 = q_start; q0 < S; q0 += 32) {
-    if (q0 + 32 < S) stage(cur ^ 1, q0 + 32);
+    if (q0 + 32 < S) stage(cur ^ 1);
 
     // S^T (A=K lds, B=imgq) and dP^T (A=V regs, B=imgdo)
     f32x16 accS = (f32x16)(0.f), accDP = (f32x16)(0.f);
@@ -639,7 +767,22 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       const int key = DROW(r, hb);
       *(short*)(my_p + key * 80 + col * 2) = f2bf(pv[r]);
     }
-    // dV[key][dv] += P(A) @ tr(imgdo)(B)
+    // dV[key][dv] += P(A) @ tr(imgdo)(B). Immediate-offset reads (one
+    // base add per t) — the dkv kernel sits at the 255-VGPR edge, so no
+    // extra pipeline slots here, but the address chains still go.
+#define DKV_TRLOOP(ACC, A0, A1, BASE)                                        \
+  do {                                                                       \
+    _Pragma("unroll") for (int t = 0; t < NT; ++t) {                         \
+      U2x64 f0, f1;                                                          \
+      const unsigned b_t = (BASE) + t * 2048;                                \
+      tr_issue4<0, 128, 512, 640>(b_t, f0, f1);                              \
+      tr_wait<0>(f0, f1);                                                    \
+      ACC[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(A0, f0.v, ACC[t],     \
+                                                       0, 0, 0);             \
+      ACC[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(A1, f1.v, ACC[t],     \
+                                                       0, 0, 0);             \
+    }                                                                        \
+  } while (0)
     {
       bf16x8v pa[2];
 #pragma unroll
@@ -647,23 +790,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         const int inrow = kc * 32 + hb * 16;
         pa[kc] = *(const bf16x8v*)(my_p + col * 80 + inrow);
       }
-#pragma unroll
-      for (int t = 0; t < NT; ++t) {
-        const int dvblk = t * 2 + (col >> 4);
-        const int qa = hb * 8, qb2 = 16 + hb * 8;
-        const int rga0 = qa >> 2, rga1 = (qa + 4) >> 2;
-        const int rgb0 = qb2 >> 2, rgb1 = (qb2 + 4) >> 2;
-        const TR4 f = tr_read4(
-            DOIMG(cur),
-            (dvblk * 8 + rga0) * 128 + (((lane & 15) * 8) ^ ((rga0 & 2) << 3)),
-            (dvblk * 8 + rga1) * 128 + (((lane & 15) * 8) ^ ((rga1 & 2) << 3)),
-            (dvblk * 8 + rgb0) * 128 + (((lane & 15) * 8) ^ ((rgb0 & 2) << 3)),
-            (dvblk * 8 + rgb1) * 128 + (((lane & 15) * 8) ^ ((rgb1 & 2) << 3)));
-        accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], f.a,
-                                                           accDV[t], 0, 0, 0);
-        accDV[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], f.b,
-                                                           accDV[t], 0, 0, 0);
-      }
+      const unsigned dob = (unsigned)(unsigned long long)DOIMG(cur) + kv_swz;
+      DKV_TRLOOP(accDV, pa[0], pa[1], dob);
     }
     // dS -> my_p, then dK[key][dk] += dS(A) @ tr(imgq)(B)
 #pragma unroll
@@ -678,24 +806,10 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         const int inrow = kc * 32 + hb * 16;
         da[kc] = *(const bf16x8v*)(my_p + col * 80 + inrow);
       }
-#pragma unroll
-      for (int t = 0; t < NT; ++t) {
-        const int dkblk = t * 2 + (col >> 4);
-        const int qa = hb * 8, qb2 = 16 + hb * 8;
-        const int rga0 = qa >> 2, rga1 = (qa + 4) >> 2;
-        const int rgb0 = qb2 >> 2, rgb1 = (qb2 + 4) >> 2;
-        const TR4 f = tr_read4(
-            QIMG(cur),
-            (dkblk * 8 + rga0) * 128 + (((lane & 15) * 8) ^ ((rga0 & 2) << 3)),
-            (dkblk * 8 + rga1) * 128 + (((lane & 15) * 8) ^ ((rga1 & 2) << 3)),
-            (dkblk * 8 + rgb0) * 128 + (((lane & 15) * 8) ^ ((rgb0 & 2) << 3)),
-            (dkblk * 8 + rgb1) * 128 + (((lane & 15) * 8) ^ ((rgb1 & 2) << 3)));
-        accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[0], f.a,
-                                                           accDK[t], 0, 0, 0);
-        accDK[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da[1], f.b,
-                                                           accDK[t], 0, 0, 0);
-      }
+      const unsigned qib = (unsigned)(unsigned long long)QIMG(cur) + kv_swz;
+      DKV_TRLOOP(accDK, da[0], da[1], qib);
     }
+#undef DKV_TRLOOP
     __syncthreads();
     cur ^= 1;
   }
@@ -738,7 +852,7 @@ void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
                      float* lse, int B, int S, int H, int KVH, int D,
                      float scale, long long vstride, hipStream_t stream) {
   dim3 grid(S / 128, B * H);
-  const int lds = 2 * (64 * D * 2 + D * 64 * 2);  // dbuf K + Vt
+  const int lds = 3 * 64 * D * 2;  // single K image + dbuf V
   if (D == 128)
     attn_fwd_kernel<128><<<grid, 256, lds, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse, B,
