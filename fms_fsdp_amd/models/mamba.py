@@ -17,6 +17,7 @@ Block layout matches mamba_ssm's Mamba2:
 """
 
 import math
+import os
 from dataclasses import dataclass, field
 from typing import List
 
@@ -198,6 +199,49 @@ class Mamba2Mixer(nn.Module):
         nn.init.ones_(self.D)
         self.norm.reset_parameters()
 
+    def _scan_fused(self, b, l, x, dt, z, B, C, dtb, Alog, D_):
+        """GPU SSD scan: the elementwise chains run as fused HIP kernels
+        (ops/hip/ssd.hip — prep/xdt/scores-decay/ygate, each with a
+        custom backward); only the MFMA-shaped work (scores, y_diag,
+        states, y_off einsums on hipBLASLt) and the tiny nc x nc
+        inter-chunk recurrence stay as torch ops. Inputs x/dt/z/B/C are
+        STRIDED slices of the fused projections — no contiguous copies.
+        Returns the gated (b, l, d_inner) bf16 activations (pre-norm)."""
+        h, p = self.nheads, self.headdim
+        g, n = self.ngroups, self.d_state
+        Q = self.chunk
+        nc = l // Q
+        rep = h // g
+        dt2 = dt.reshape(b * l, h)
+        x2 = x.reshape(b * l, h * p)
+        z2 = z.reshape(b * l, h * p)
+        dtf, dacs = ops.ssd_prep(dt2, dtb, Alog, Q)        # (b*nc*h, Q) fp32
+        xdt, xdtd = ops.ssd_xdt(x2, dtf, dacs, h, p, Q)    # bf16 pair
+        Bm = B.reshape(b, nc, Q, g, n)
+        Cm = C.reshape(b, nc, Q, g, n)
+        # scores per GROUP (g << h), decay folded in-kernel (L never
+        # materialized; d_scores comes back from the fused backward)
+        scores = torch.einsum("bcqgn,bckgn->bcgqk", Cm, Bm).contiguous()
+        sL = ops.ssd_scores_decay(dacs, scores.view(b * nc * g, Q, Q), h, g)
+        y_diag = torch.einsum(
+            "bcgrqk,bckgrp->bcqgrp", sL.view(b, nc, g, rep, Q, Q),
+            xdt.view(b, nc, Q, g, rep, p))                 # bf16
+        states = torch.einsum(
+            "bckgn,bckgrp->bcgrnp", Bm,
+            xdtd.view(b, nc, Q, g, rep, p)).float().reshape(b, nc, h, n, p)
+        # inter-chunk recurrence in closed form (see ssd_chunked)
+        G = dacs.view(b, nc, h, Q)[..., -1].permute(0, 2, 1)
+        W = torch.exp(segsum(G))
+        P_ = torch.einsum("bhzc,bchnp->bzhnp", W, states)
+        prev = torch.cat([torch.zeros_like(P_[:, :1]), P_[:, :-1]], dim=1)
+        y_off = torch.einsum(
+            "bcqgn,bcgrnp->bcqgrp", Cm,
+            prev.view(b, nc, g, rep, n, p).to(torch.bfloat16))
+        out = ops.ssd_ygate(
+            y_diag.reshape(b * l, h * p), y_off.reshape(b * l, h * p),
+            dacs, x2, D_, z2, h, p, Q)
+        return out.view(b, l, h * p)
+
     def forward(self, u):
         b, l, _ = u.shape
         zxbcdt = ops.linear_flat(u, self.in_proj.weight)
@@ -211,6 +255,18 @@ class Mamba2Mixer(nn.Module):
             xBC, [self.d_inner,
                   self.ngroups * self.d_state, self.ngroups * self.d_state],
             dim=-1)
+
+        if u.is_cuda and u.dtype == torch.bfloat16 and l % self.chunk == 0 \
+                and not getattr(self, "_force_torch_scan", False) \
+                and os.environ.get("FMS_AMD_FORCE_TORCH_SCAN") != "1":
+            args = (b, l, x, dt, z, B, C, self.dt_bias, self.A_log, self.D)
+            if torch.is_grad_enabled() and self.training:
+                y = torch.utils.checkpoint.checkpoint(
+                    self._scan_fused, *args, use_reentrant=False)
+            else:
+                y = self._scan_fused(*args)
+            y = self.norm(y)
+            return ops.linear_flat(y, self.out_proj.weight)
 
         def _scan(x_, dt_, dtb, Alog, B_, C_, D_):
             # fp32 SSD; recomputed in backward (the chunked intermediates

@@ -484,6 +484,128 @@ def segsum_exp(cs):
 
 
 # --------------------------------------------------------------------------
+# Fused SSD scan pieces (Mamba2): every fp32 elementwise chain around the
+# batched GEMMs is a HIP kernel with a custom backward (north-star item:
+# hand-written selective-scan; reference reaches mamba_ssm's kernels from
+# main_training_mamba.py:8-9,67). GPU-only: the CPU path keeps the plain
+# torch ssd_chunked as the numerics oracle.
+# --------------------------------------------------------------------------
+class _SSDPrepFn(torch.autograd.Function):
+    """(dt2d strided bf16 (b*l, H), dt_bias, A_log) ->
+    dtf (N,Q) fp32, dacs (N,Q) fp32 with N=(b*l/Q)*H, n = bc*H + h:
+    dtf = softplus(dt + bias); dacs = chunk-cumsum(dtf * -exp(A_log))."""
+
+    @staticmethod
+    def forward(ctx, dt2d, dt_bias, A_log, chunk):
+        ext = _require_ext("ssd_prep")
+        bias_f = dt_bias.detach().float().contiguous()
+        alog_f = A_log.detach().float().contiguous()
+        dtf, dacs = ext.ssd_prep_fwd(dt2d, bias_f, alog_f, chunk)
+        ctx.save_for_backward(dt2d, bias_f, alog_f)
+        ctx.meta = (chunk, dt_bias.dtype, A_log.dtype)
+        return dtf, dacs
+
+    @staticmethod
+    def backward(ctx, ddtf, ddacs):
+        dt2d, bias_f, alog_f = ctx.saved_tensors
+        chunk, bdt, adt = ctx.meta
+        ddt, dbias, dalog = _C.ssd_prep_bwd(
+            ddtf.contiguous(), ddacs.contiguous(), dt2d, bias_f, alog_f,
+            chunk)
+        return ddt, dbias.to(bdt), dalog.to(adt), None
+
+
+def ssd_prep(dt2d, dt_bias, A_log, chunk):
+    return _SSDPrepFn.apply(dt2d, dt_bias, A_log, chunk)
+
+
+class _SSDXdtFn(torch.autograd.Function):
+    """x2d (b*l, H*P) strided bf16 -> (xdt, xdt_decayed) bf16 contiguous
+    (layout (b,nc,Q,H,P)); xdt = x*dtf, decayed by exp(dacs_end-dacs)."""
+
+    @staticmethod
+    def forward(ctx, x2d, dtf, dacs, H, P, Q):
+        ext = _require_ext("ssd_xdt")
+        xdt, xdtd = ext.ssd_xdt_fwd(x2d, dtf, dacs, H, P, Q)
+        ctx.save_for_backward(x2d, dtf, dacs)
+        ctx.meta = (H, P, Q)
+        return xdt, xdtd
+
+    @staticmethod
+    def backward(ctx, dxdt, dxdtd):
+        x2d, dtf, dacs = ctx.saved_tensors
+        H, P, Q = ctx.meta
+        dx, ddtf, sdec = _C.ssd_xdt_bwd(
+            dxdt.contiguous(), dxdtd.contiguous(), x2d, dtf, dacs, H, P, Q)
+        # fold the decay-exponent grads: d/d dacs[q] = -sdec[q], and the
+        # end column accumulates all q of its row
+        ddacs = sdec.neg()
+        ddacs[:, -1] += sdec.sum(-1)
+        return dx, ddtf, ddacs, None, None, None
+
+
+def ssd_xdt(x2d, dtf, dacs, H, P, Q):
+    return _SSDXdtFn.apply(x2d, dtf, dacs, H, P, Q)
+
+
+class _SSDScoresDecayFn(torch.autograd.Function):
+    """sL[n,i,j] = scores[m,i,j] * exp(dacs[n,i]-dacs[n,j]) (j<=i), the
+    decay matrix L never materialized. scores per GROUP (M=b*nc*G rows);
+    backward returns per-head d_scores summed over each group's heads."""
+
+    @staticmethod
+    def forward(ctx, dacs, scores3d, H, G):
+        ext = _require_ext("ssd_sl")
+        sL = ext.ssd_sl_fwd(dacs, scores3d, H, G)
+        ctx.save_for_backward(dacs, scores3d)
+        ctx.meta = (H, G)
+        return sL
+
+    @staticmethod
+    def backward(ctx, g):
+        dacs, scores3d = ctx.saved_tensors
+        H, G = ctx.meta
+        dsh, dcs = _C.ssd_sl_bwd(g.contiguous(), scores3d, dacs, H, G)
+        rep = H // G
+        M = scores3d.shape[0]
+        Q = scores3d.shape[1]
+        dscores = dsh.view(M // G, G, rep, Q, Q) \
+            .sum(dim=2, dtype=torch.float32).to(scores3d.dtype).view(M, Q, Q)
+        return dcs, dscores, None, None
+
+
+def ssd_scores_decay(dacs, scores3d, H, G):
+    return _SSDScoresDecayFn.apply(dacs, scores3d, H, G)
+
+
+class _SSDYGateFn(torch.autograd.Function):
+    """out = (ydiag + yoff*exp(dacs) + x*D) * silu(z), bf16 in one pass
+    (y assembly + D residual + the gated epilogue)."""
+
+    @staticmethod
+    def forward(ctx, ydiag, yoff, dacs, x2d, D, z2d, H, P, Q):
+        ext = _require_ext("ssd_ygate")
+        D_f = D.detach().float().contiguous()
+        out = ext.ssd_ygate_fwd(ydiag, yoff, dacs, x2d, D_f, z2d, H, P, Q)
+        ctx.save_for_backward(ydiag, yoff, dacs, x2d, D_f, z2d)
+        ctx.meta = (H, P, Q, D.dtype)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ydiag, yoff, dacs, x2d, D_f, z2d = ctx.saved_tensors
+        H, P, Q, ddt = ctx.meta
+        dydiag, dyoff, ddacs, dx, dD_rows, dz = _C.ssd_ygate_bwd(
+            dout.contiguous(), ydiag, yoff, dacs, x2d, D_f, z2d, H, P, Q)
+        dD = dD_rows.sum(0)
+        return (dydiag, dyoff, ddacs, dx, dD.to(ddt), dz, None, None, None)
+
+
+def ssd_ygate(ydiag, yoff, dacs, x2d, D, z2d, H, P, Q):
+    return _SSDYGateFn.apply(ydiag, yoff, dacs, x2d, D, z2d, H, P, Q)
+
+
+# --------------------------------------------------------------------------
 # Fused AdamW on flat fp32 shards + multi-tensor sq-norm
 # --------------------------------------------------------------------------
 def fused_adamw(p, g, m, v, step, lr, beta1, beta2, eps, weight_decay,

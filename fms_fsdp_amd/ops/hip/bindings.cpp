@@ -5,6 +5,7 @@
 #include <hip/hip_runtime_api.h>
 
 #include <tuple>
+#include <vector>
 
 using torch::Tensor;
 
@@ -40,6 +41,29 @@ void launch_cconv_bwd(const void*, const void*, const void*, const float*,
 void launch_segsum_exp_fwd(const float*, void*, long long, int, hipStream_t);
 void launch_segsum_exp_bwd(const void*, const float*, float*, long long, int,
                            hipStream_t);
+void launch_ssd_prep_fwd(const void*, const float*, const float*, float*,
+                         float*, long long, int, int, long long, hipStream_t);
+void launch_ssd_prep_bwd(const float*, const float*, const void*,
+                         const float*, const float*, void*, float*, float*,
+                         long long, int, int, long long, long long,
+                         hipStream_t);
+void launch_ssd_xdt_fwd(const void*, const float*, const float*, void*, void*,
+                        long long, int, int, int, long long, hipStream_t);
+void launch_ssd_xdt_bwd(const void*, const void*, const void*, const float*,
+                        const float*, void*, float*, float*, long long, int,
+                        int, int, long long, long long, hipStream_t);
+void launch_ssd_sl_fwd(const float*, const void*, void*, long long, int, int,
+                       int, hipStream_t);
+void launch_ssd_sl_bwd(const void*, const void*, const float*, void*, float*,
+                       long long, int, int, int, hipStream_t);
+void launch_ssd_ygate_fwd(const void*, const void*, const float*, const void*,
+                          const float*, const void*, void*, long long, int,
+                          int, int, long long, long long, hipStream_t);
+void launch_ssd_ygate_bwd(const void*, const void*, const void*, const float*,
+                          const void*, const float*, const void*, void*,
+                          void*, float*, void*, float*, void*, long long, int,
+                          int, int, long long, long long, long long,
+                          long long, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -313,6 +337,145 @@ Tensor segsum_exp_bwd(Tensor g, Tensor cs) {
   return dcs;
 }
 
+
+// ---- fused SSD scan pieces (see ops/hip/ssd.hip) ----
+// strided 2-D bf16 slice (rows, cols) with unit inner stride
+static long long slice_stride(const Tensor& t, const char* nm) {
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, nm, " must be bf16");
+  TORCH_CHECK(t.dim() == 2 && t.stride(1) == 1, nm, " must be 2-D row-major");
+  return t.stride(0);
+}
+
+std::tuple<Tensor, Tensor> ssd_prep_fwd(Tensor dt, Tensor bias, Tensor alog,
+                                        int64_t Q) {
+  const long long sdt = slice_stride(dt, "dt");
+  const long long rows = dt.size(0);
+  const int H = dt.size(1);
+  TORCH_CHECK(rows % Q == 0 && Q <= 256 && Q % 64 == 0);
+  const long long N = rows / Q * H;
+  auto opt = dt.options().dtype(torch::kFloat32);
+  auto dtf = torch::empty({N, Q}, opt);
+  auto dacs = torch::empty({N, Q}, opt);
+  launch_ssd_prep_fwd(dt.data_ptr(), bias.data_ptr<float>(),
+                      alog.data_ptr<float>(), dtf.data_ptr<float>(),
+                      dacs.data_ptr<float>(), N, H, (int)Q, sdt,
+                      cur_stream());
+  return {dtf, dacs};
+}
+
+std::tuple<Tensor, Tensor, Tensor> ssd_prep_bwd(Tensor ddtf, Tensor ddacs,
+                                                Tensor dt, Tensor bias,
+                                                Tensor alog, int64_t Q) {
+  const long long sdt = slice_stride(dt, "dt");
+  const long long rows = dt.size(0);
+  const int H = dt.size(1);
+  const long long N = rows / Q * H;
+  auto ddt = torch::empty({rows, (long long)H},
+                          dt.options().dtype(torch::kBFloat16));
+  auto dbias = torch::zeros({H}, dt.options().dtype(torch::kFloat32));
+  auto dalog = torch::zeros({H}, dt.options().dtype(torch::kFloat32));
+  launch_ssd_prep_bwd(ddtf.data_ptr<float>(), ddacs.data_ptr<float>(),
+                      dt.data_ptr(), bias.data_ptr<float>(),
+                      alog.data_ptr<float>(), ddt.data_ptr(),
+                      dbias.data_ptr<float>(), dalog.data_ptr<float>(), N, H,
+                      (int)Q, sdt, H, cur_stream());
+  return {ddt, dbias, dalog};
+}
+
+std::tuple<Tensor, Tensor> ssd_xdt_fwd(Tensor x, Tensor dtf, Tensor dacs,
+                                       int64_t H, int64_t P, int64_t Q) {
+  const long long sx = slice_stride(x, "x");
+  const long long rows = x.size(0);
+  auto xdt = torch::empty({rows, H * P},
+                          x.options().dtype(torch::kBFloat16));
+  auto xdtd = torch::empty_like(xdt);
+  const long long total8 = rows * H * P / 8;
+  launch_ssd_xdt_fwd(x.data_ptr(), dtf.data_ptr<float>(),
+                     dacs.data_ptr<float>(), xdt.data_ptr(), xdtd.data_ptr(),
+                     total8, (int)H, (int)Q, (int)P, sx, cur_stream());
+  return {xdt, xdtd};
+}
+
+std::tuple<Tensor, Tensor, Tensor> ssd_xdt_bwd(Tensor dxdt, Tensor dxdtd,
+                                               Tensor x, Tensor dtf,
+                                               Tensor dacs, int64_t H,
+                                               int64_t P, int64_t Q) {
+  const long long sx = slice_stride(x, "x");
+  const long long rows = x.size(0);
+  auto dx = torch::empty({rows, H * P}, x.options().dtype(torch::kBFloat16));
+  auto ddtf = torch::empty_like(dtf);
+  auto sdec = torch::empty_like(dacs);
+  launch_ssd_xdt_bwd(dxdt.data_ptr(), dxdtd.data_ptr(), x.data_ptr(),
+                     dtf.data_ptr<float>(), dacs.data_ptr<float>(),
+                     dx.data_ptr(), ddtf.data_ptr<float>(),
+                     sdec.data_ptr<float>(), rows * H, (int)H, (int)Q,
+                     (int)P, sx, H * P, cur_stream());
+  return {dx, ddtf, sdec};
+}
+
+Tensor ssd_sl_fwd(Tensor dacs, Tensor scores, int64_t H, int64_t G) {
+  TORCH_CHECK(scores.scalar_type() == torch::kBFloat16 &&
+              scores.is_contiguous());
+  const long long N = dacs.size(0);
+  const int Q = dacs.size(1);
+  auto out = torch::empty({N, (long long)Q, (long long)Q},
+                          scores.options());
+  launch_ssd_sl_fwd(dacs.data_ptr<float>(), scores.data_ptr(),
+                    out.data_ptr(), N * Q * (Q / 8), (int)H, (int)G, Q,
+                    cur_stream());
+  return out;
+}
+
+std::tuple<Tensor, Tensor> ssd_sl_bwd(Tensor g, Tensor scores, Tensor dacs,
+                                      int64_t H, int64_t G) {
+  const long long N = dacs.size(0);
+  const int Q = dacs.size(1);
+  auto dsh = torch::empty({N, (long long)Q, (long long)Q}, g.options());
+  auto dcs = torch::zeros_like(dacs);
+  launch_ssd_sl_bwd(g.data_ptr(), scores.data_ptr(), dacs.data_ptr<float>(),
+                    dsh.data_ptr(), dcs.data_ptr<float>(), N, (int)H, (int)G,
+                    Q, cur_stream());
+  return {dsh, dcs};
+}
+
+Tensor ssd_ygate_fwd(Tensor ydiag, Tensor yoff, Tensor dacs, Tensor x,
+                     Tensor Dp, Tensor z, int64_t H, int64_t P, int64_t Q) {
+  const long long sx = slice_stride(x, "x");
+  const long long sz = slice_stride(z, "z");
+  TORCH_CHECK(ydiag.is_contiguous() && yoff.is_contiguous());
+  const long long rows = x.size(0);
+  auto out = torch::empty({rows, H * P}, x.options().dtype(torch::kBFloat16));
+  launch_ssd_ygate_fwd(ydiag.data_ptr(), yoff.data_ptr(),
+                       dacs.data_ptr<float>(), x.data_ptr(),
+                       Dp.data_ptr<float>(), z.data_ptr(), out.data_ptr(),
+                       rows * H * P / 8, (int)H, (int)Q, (int)P, sx, sz,
+                       cur_stream());
+  return out;
+}
+
+std::vector<Tensor> ssd_ygate_bwd(Tensor dout, Tensor ydiag, Tensor yoff,
+                                  Tensor dacs, Tensor x, Tensor Dp, Tensor z,
+                                  int64_t H, int64_t P, int64_t Q) {
+  const long long sx = slice_stride(x, "x");
+  const long long sz = slice_stride(z, "z");
+  const long long rows = x.size(0);
+  auto o = x.options().dtype(torch::kBFloat16);
+  auto dydiag = torch::empty({rows, H * P}, o);
+  auto dyoff = torch::empty({rows, H * P}, o);
+  auto ddacs = torch::empty_like(dacs);
+  auto dx = torch::empty({rows, H * P}, o);
+  auto dD_rows = torch::empty({rows, H}, x.options().dtype(torch::kFloat32));
+  auto dz = torch::empty({rows, H * P}, o);
+  launch_ssd_ygate_bwd(dout.data_ptr(), ydiag.data_ptr(), yoff.data_ptr(),
+                       dacs.data_ptr<float>(), x.data_ptr(),
+                       Dp.data_ptr<float>(), z.data_ptr(), dydiag.data_ptr(),
+                       dyoff.data_ptr(), ddacs.data_ptr<float>(),
+                       dx.data_ptr(), dD_rows.data_ptr<float>(),
+                       dz.data_ptr(), rows * H, (int)H, (int)Q, (int)P, sx,
+                       sz, H * P, H * P, cur_stream());
+  return {dydiag, dyoff, ddacs, dx, dD_rows, dz};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("segsum_exp_fwd", &segsum_exp_fwd);
   mod.def("segsum_exp_bwd", &segsum_exp_bwd);
@@ -328,6 +491,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_fwd_bwd", &ce_fwd_bwd);
   mod.def("adamw", &adamw);
   mod.def("sq_norm_accum", &sq_norm_accum);
+  mod.def("ssd_prep_fwd", &ssd_prep_fwd);
+  mod.def("ssd_prep_bwd", &ssd_prep_bwd);
+  mod.def("ssd_xdt_fwd", &ssd_xdt_fwd);
+  mod.def("ssd_xdt_bwd", &ssd_xdt_bwd);
+  mod.def("ssd_sl_fwd", &ssd_sl_fwd);
+  mod.def("ssd_sl_bwd", &ssd_sl_bwd);
+  mod.def("ssd_ygate_fwd", &ssd_ygate_fwd);
+  mod.def("ssd_ygate_bwd", &ssd_ygate_bwd);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_bwd", &attn_bwd);
 }
