@@ -86,13 +86,16 @@ def main(**kwargs):
                   f"running {arch} replicated")
         cfg.sharding_strategy = "fsdp"
     if cfg.sharding_strategy == "tp" and world_size > 1:
+        # 2-D dp x tp device mesh for the frozen base model (reference:
+        # train_speculator.py:128-142 builds base_model_mesh ("dp","tp")
+        # plus a flat 1-D speculator mesh; our speculator uses the world
+        # group via ShardedModel "ddp", which IS that flat mesh)
         tp_size = min(cfg.tp_size, world_size)
-        n_dp = world_size // tp_size
-        for d in range(n_dp):
-            ranks = list(range(d * tp_size, (d + 1) * tp_size))
-            g = dist.new_group(ranks)
-            if rank in ranks:
-                tp_group = g
+        from torch.distributed.device_mesh import init_device_mesh
+        base_model_mesh = init_device_mesh(
+            "cuda" if torch.cuda.is_available() else "cpu",
+            (world_size // tp_size, tp_size), mesh_dim_names=("dp", "tp"))
+        tp_group = base_model_mesh["tp"].get_group()
         model = tp_shard_llama(model, tp_group)
     elif cfg.sharding_strategy in ("fsdp", "hsdp") and world_size > 1:
         # frozen replicated weights would also work; keep them whole per

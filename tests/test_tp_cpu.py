@@ -126,3 +126,41 @@ def test_tp_speculator_training_loop(tmp_path):
         res[tag] = (sums, maxes)
     assert res["ok0"] == res["ok1"], "speculator diverged across dp ranks"
     assert all(abs(x) > 0 for x in res["ok0"][1]), "speculator never trained"
+
+
+def _entry_worker(rank, world, tmpdir, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29697"
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from speculator import train_speculator as ts
+        ts.main(model_variant="llama2_125m", use_dummy_dataset=True,
+                batch_size=1, seq_length=128, num_steps=2,
+                report_interval=1, checkpoint_interval=100,
+                mixed_precision=False, n_speculator_heads=2,
+                speculator_width=64, stage2_start_step=10,
+                model_path="/nonexistent", ckpt_save_path=tmpdir,
+                ckpt_load_path=tmpdir, vocab_size=256, learning_rate=1e-4,
+                sharding_strategy="tp", tp_size=2)
+        if rank == 0:
+            q.put(("ok", None))
+    except Exception as e:
+        q.put(("err", f"{type(e).__name__}: {e}"))
+        raise
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_speculator_entry_tp_world2(tmp_path):
+    """The full train_speculator entry under sharding_strategy=tp at
+    world 2: dp x tp device-mesh construction (reference
+    train_speculator.py:128-142), tp_shard_llama, and the TP loop."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_entry_worker, args=(2, str(tmp_path), q), nprocs=2, join=True)
+    status, err = q.get()
+    assert status == "ok", err
