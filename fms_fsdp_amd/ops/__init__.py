@@ -186,6 +186,11 @@ class _DirectWgradLinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         dy2 = dy.reshape(-1, dy.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
+        # NOTE: rerouting this dgrad through a transposed weight copy
+        # (mm(dy, wT.t()) = hipBLASLt's faster "NN" class instead of the
+        # "NT" class this layout hits) measured 29 ms/step SLOWER on the
+        # 7B bench — the .t().contiguous() sweeps cost more than the
+        # GEMM class difference saves. Measured and rejected.
         dx = (dy2 @ w).view(x.shape)
         # wgrad straight into the flat grad buffer. First touch since
         # zero_grad overwrites (beta=0) so the runtime never has to
@@ -397,6 +402,7 @@ class _LinearCEFn(torch.autograd.Function):
         loss_sum = torch.zeros((), device=x.device, dtype=torch.float32)
         count = (lab != ignore_index).sum()
         denom = count.clamp(min=1).float()
+
         for i in range(0, n, _LinearCEFn.CHUNK):
             xc = x2d[i:i + _LinearCEFn.CHUNK]
             lc = lab[i:i + _LinearCEFn.CHUNK]
