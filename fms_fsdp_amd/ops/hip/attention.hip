@@ -634,7 +634,6 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     void* __restrict__ dkg, void* __restrict__ dvg, int B, int S, int H,
     int KVH, float scale, long long vstride, long long dvstride) {
   constexpr int KVB = 32;   // keys per wave; block = 4 waves = 128 keys
-  constexpr int KSWZ = (D == 128) ? 15 : 7;
   constexpr int NC = D / 16;
   constexpr int NT = D / 32;
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -672,16 +671,17 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   char* my_k = k_lds + wid * KVB * D * 2;
   char* my_p = p_lds + wid * KVB * 80;
 
-  // stage this wave's K rows once (swizzled row-major)
+  // stage this wave's K rows once, in the SUBTILED image format (same
+  // as the Q/dO images): the S-phase plain reads hit the proven
+  // conflict-free pattern (PMC: the old row-swizzled tile carried the
+  // residual bank conflicts)
   {
-    constexpr int BPR = D * 2;
-    constexpr int CHUNKS = KVB * BPR / 16;
+    constexpr int CHUNKS = KVB * D / 8;     // 16 B chunks
     for (int i = lane; i < CHUNKS; i += 64) {
-      const int row = i / (BPR / 16);
-      const int cb = (i % (BPR / 16)) * 16;
-      const long long g = (long long)(kv0 + row) * krow_stride + cb / 2;
-      *(f32x4*)(my_k + ((row * BPR + cb) ^ ((row & KSWZ) << 4))) =
-          *(const f32x4*)(kbase + g);
+      const int row = i / (D / 8);
+      const int cb = (i % (D / 8)) * 8;
+      const long long g = (long long)(kv0 + row) * krow_stride + cb;
+      *(f32x4*)(my_k + SUBT_OFF(row, cb, 8)) = *(const f32x4*)(kbase + g);
     }
   }
   // V fragments stay in registers (A-operand rows = this lane's key)
@@ -736,10 +736,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     f32x16 accS = (f32x16)(0.f), accDP = (f32x16)(0.f);
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
-      const int inrow = c * 32 + hb * 16;
-      const bf16x8v ka =
-          *(const bf16x8v*)(my_k + ((col * (D * 2) + inrow) ^ ((col & KSWZ) << 4)));
       const int boff = SUBT_OFF(col, c * 16 + hb * 8, 8);
+      const bf16x8v ka = *(const bf16x8v*)(my_k + boff);
       const bf16x8v qbf = *(const bf16x8v*)(QIMG(cur) + boff);
       const bf16x8v dbf = *(const bf16x8v*)(DOIMG(cur) + boff);
       accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qbf, accS, 0, 0, 0);
