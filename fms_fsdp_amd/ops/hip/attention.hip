@@ -206,22 +206,30 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_kernel(
   // v may be a strided view (a slice of the fused qkv projection)
   const short* vbase = vg + (long long)b * S * vstride + (long long)kvh * D;
 
-  // Q -> B-fragments (registers, reused all tiles)
+  // softmax runs in base 2 (v_exp_f32 is natively 2^x; folding log2(e)
+  // into the scale drops one v_mul per exponential)
+  const float scale2 = scale * 1.4426950408889634f;
+  // Q -> B-fragments (registers, reused all tiles), PRE-SCALED by
+  // scale*log2e: accS arrives already in softmax units, dropping 16
+  // VALU muls per sub-tile (PMC: every kernel is VALU-bound). The bf16
+  // rounding of q*scale2 is within the operands' own precision.
   bf16x8v qb[NC];
   {
     const short* qp = qbase + (long long)my_q * qrow_stride + hb * 8;
 #pragma unroll
-    for (int c = 0; c < NC; ++c)
-      qb[c] = *(const bf16x8v*)(qp + c * 16);
+    for (int c = 0; c < NC; ++c) {
+      bf16x8v qv = *(const bf16x8v*)(qp + c * 16);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        qv[e] = (__bf16)((float)qv[e] * scale2);
+      qb[c] = qv;
+    }
   }
 
   f32x16 accO[NT];
 #pragma unroll
   for (int t = 0; t < NT; ++t) accO[t] = (f32x16)(0.f);
   float m_run = -1e30f, l_run = 0.f;
-  // softmax runs in base 2 (v_exp_f32 is natively 2^x; folding log2(e)
-  // into the scale drops one v_mul per exponential)
-  const float scale2 = scale * 1.4426950408889634f;
 
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
   const int t256 = threadIdx.x;
@@ -297,7 +305,7 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_kernel(
       float mt = -1e30f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        float sv = accS[r] * scale2;
+        float sv = accS[r];                   // q pre-scaled at load
         if (partial && (kv32 + DROW(r, hb)) > my_q) sv = -1e30f;
         p[r] = sv;
         mt = fmaxf(mt, sv);
@@ -487,20 +495,24 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   const short* kbase = kg + ((long long)b * S * KVH + (long long)kvh) * D;
   const short* vbase = vg + (long long)b * S * vstride + (long long)kvh * D;
 
+  const float scale2 = scale * 1.4426950408889634f;
   bf16x8v qb[NC], dob[NC];
   {
     const short* qp = qbase + (long long)my_q * qrow_stride + hb * 8;
     const short* dp = dobase + (long long)my_q * qrow_stride + hb * 8;
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
-      qb[c] = *(const bf16x8v*)(qp + c * 16);
+      bf16x8v qv = *(const bf16x8v*)(qp + c * 16);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)   // pre-scale (see fwd kernel)
+        qv[e] = (__bf16)((float)qv[e] * scale2);
+      qb[c] = qv;
       dob[c] = *(const bf16x8v*)(dp + c * 16);
     }
   }
   const float my_lse2 =
       lseg[((long long)bh) * S + my_q] * 1.4426950408889634f;
   const float my_delta = deltag[((long long)bh) * S + my_q];
-  const float scale2 = scale * 1.4426950408889634f;
 
   f32x16 accDQ[NT];
 #pragma unroll
@@ -559,7 +571,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int key = kv0 + DROW(r, hb);
-      const float pval = exp2f(accS[r] * scale2 - my_lse2);
+      const float pval = exp2f(accS[r] - my_lse2);  // q pre-scaled
       float v = pval * (accDP[r] - my_delta) * scale;
       if (key > my_q) v = 0.f;
       ds[r] = v;
