@@ -1,13 +1,15 @@
 // Fused causal depthwise conv1d + silu (mamba xBC conv, width<=4) and its
 // backward. Memory-bound; bf16x8 vectorized channel access (guide G13).
 // Replaces mamba_ssm's causal-conv1d .cu kernel.
-// x (b, l, C) bf16 row-major; weight (C, W) bf16; bias (C) fp32-or-bf16.
+// x (b, l, C) bf16 row-major; weight (C, W) bf16; bias (C) fp32.
 //
-// The W=4 path (every registry config) loads each thread's 8-channel
-// weight block as FOUR bf16x8 vectors and the bias as two f32x4 — the
-// original scalar per-tap loads (32 x 2B per thread per row) were
-// issue-bound and ran ~12x off the HBM roofline (torch.profiler:
-// ~300us per kernel for ~180 MB of traffic).
+// STRIP kernels: one thread produces 8 consecutive timesteps for its
+// 8-channel block, with the x (or g) rows entering a W-deep sliding
+// register window — each row is loaded ONCE and the weights are
+// unpacked once per strip. The per-(t,c) predecessor re-read every row
+// W times and re-unpacked the weights per output; it measured ~1.9 TB/s
+// effective (230-300 us per call, 12% of the mamba step across the four
+// kernels).
 #include "common.h"
 
 // per-thread 8-channel weight block: channels c0..c0+7, W taps each,
@@ -27,42 +29,78 @@ struct WBlock {
 };
 
 template <int W>
+__device__ __forceinline__ void unpack_wb(const short* w, int c0,
+                                          float wf[8][W]) {
+  WBlock<W> wb;
+  wb.load(w, c0);
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+#pragma unroll
+    for (int wi = 0; wi < W; ++wi) wf[j][wi] = wb.at(j, wi);
+}
+
+__device__ __forceinline__ void load_row8(const short* p, float o[8]) {
+  const bf16x8 v = *(const bf16x8*)p;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) o[j] = bf2f(v.v[j]);
+}
+
+// strip index decode: i -> (b, ts, c0)
+#define STRIP_DECODE()                                                     \
+  const int C8 = C / 8;                                                    \
+  const int LS = (L + 7) / 8;                                              \
+  const int c0 = (int)(i % C8) * 8;                                        \
+  long long r_ = i / C8;                                                   \
+  const int ts = (int)(r_ % LS) * 8;                                       \
+  const long long b_ = r_ / LS;
+
+template <int W>
 __global__ void cconv_fwd_kernel(const short* __restrict__ x,
                                  const short* __restrict__ w,
                                  const float* __restrict__ bias,
                                  short* __restrict__ y,
-                                 int L, int C, long long total8) {
+                                 int L, int C, long long nstrip) {
   const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= total8) return;
-  const int C8 = C / 8;
-  const long long bt = i / C8;          // (b, t) flattened
-  const int c0 = (int)(i % C8) * 8;
-  const int t = (int)(bt % L);
-  const long long row0 = bt - t;        // start of this sequence
-
-  WBlock<W> wb;
-  wb.load(w, c0);
-  const f32x4 b0 = *(const f32x4*)(bias + c0);
-  const f32x4 b1 = *(const f32x4*)(bias + c0 + 4);
-  float acc[8];
+  if (i >= nstrip) return;
+  STRIP_DECODE();
+  const short* xb = x + b_ * L * (long long)C + c0;
+  short* yb = y + b_ * L * (long long)C + c0;
+  float wf[8][W];
+  unpack_wb<W>(w, c0, wf);
+  const f32x4 bv0 = *(const f32x4*)(bias + c0);
+  const f32x4 bv1 = *(const f32x4*)(bias + c0 + 4);
+  float bb[8];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) acc[j] = j < 4 ? b0.v[j] : b1.v[j - 4];
+  for (int j = 0; j < 8; ++j) bb[j] = j < 4 ? bv0.v[j] : bv1.v[j - 4];
+  // ring[p] holds row ts + p - (W-1) at first; slot of row t is
+  // (t - ts + W - 1) % W (constant-folded under the unrolled tt loop)
+  float ring[W][8];
 #pragma unroll
-  for (int wi = 0; wi < W; ++wi) {
-    const int ti = t - W + 1 + wi;
-    if (ti < 0) continue;
-    const bf16x8 xv = *(const bf16x8*)(x + (row0 + ti) * C + c0);
+  for (int k = 0; k < W - 1; ++k) {
+    const int ti = ts - (W - 1) + k;
+    if (ti >= 0)
+      load_row8(xb + (long long)ti * C, ring[k]);
+    else
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      acc[j] += bf2f(xv.v[j]) * wb.at(j, wi);
+      for (int j = 0; j < 8; ++j) ring[k][j] = 0.f;
   }
-  bf16x8 o;
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    const float s = 1.f / (1.f + __expf(-acc[j]));
-    o.v[j] = f2bf(acc[j] * s);
+  for (int tt = 0; tt < 8; ++tt) {
+    const int t = ts + tt;
+    if (t >= L) break;
+    load_row8(xb + (long long)t * C, ring[(tt + W - 1) % W]);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float acc = bb[j];
+#pragma unroll
+      for (int wi = 0; wi < W; ++wi)   // row t-(W-1)+wi = slot (tt+wi)%W
+        acc += wf[j][wi] * ring[(tt + wi) % W][j];
+      const float s = 1.f / (1.f + __expf(-acc));
+      o.v[j] = f2bf(acc * s);
+    }
+    *(bf16x8*)(yb + (long long)t * C) = o;
   }
-  *(bf16x8*)(y + bt * C + c0) = o;
 }
 
 // pass 1: g[t,c] = dy[t,c] * dsilu(z[t,c]) with z recomputed
@@ -72,98 +110,157 @@ __global__ void cconv_bwd_g_kernel(const short* __restrict__ dy,
                                    const short* __restrict__ w,
                                    const float* __restrict__ bias,
                                    short* __restrict__ g,
-                                   int L, int C, long long total8) {
+                                   int L, int C, long long nstrip) {
   const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= total8) return;
-  const int C8 = C / 8;
-  const long long bt = i / C8;
-  const int c0 = (int)(i % C8) * 8;
-  const int t = (int)(bt % L);
-  const long long row0 = bt - t;
-  WBlock<W> wb;
-  wb.load(w, c0);
-  const f32x4 b0 = *(const f32x4*)(bias + c0);
-  const f32x4 b1 = *(const f32x4*)(bias + c0 + 4);
-  float acc[8];
+  if (i >= nstrip) return;
+  STRIP_DECODE();
+  const short* xb = x + b_ * L * (long long)C + c0;
+  const short* dyb = dy + b_ * L * (long long)C + c0;
+  short* gb = g + b_ * L * (long long)C + c0;
+  float wf[8][W];
+  unpack_wb<W>(w, c0, wf);
+  const f32x4 bv0 = *(const f32x4*)(bias + c0);
+  const f32x4 bv1 = *(const f32x4*)(bias + c0 + 4);
+  float bb[8];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) acc[j] = j < 4 ? b0.v[j] : b1.v[j - 4];
+  for (int j = 0; j < 8; ++j) bb[j] = j < 4 ? bv0.v[j] : bv1.v[j - 4];
+  float ring[W][8];
 #pragma unroll
-  for (int wi = 0; wi < W; ++wi) {
-    const int ti = t - W + 1 + wi;
-    if (ti < 0) continue;
-    const bf16x8 xv = *(const bf16x8*)(x + (row0 + ti) * C + c0);
+  for (int k = 0; k < W - 1; ++k) {
+    const int ti = ts - (W - 1) + k;
+    if (ti >= 0)
+      load_row8(xb + (long long)ti * C, ring[k]);
+    else
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      acc[j] += bf2f(xv.v[j]) * wb.at(j, wi);
+      for (int j = 0; j < 8; ++j) ring[k][j] = 0.f;
   }
-  const bf16x8 d = *(const bf16x8*)(dy + bt * C + c0);
-  bf16x8 o;
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    const float z = acc[j];
-    const float s = 1.f / (1.f + __expf(-z));
-    o.v[j] = f2bf(bf2f(d.v[j]) * s * (1.f + z * (1.f - s)));
+  for (int tt = 0; tt < 8; ++tt) {
+    const int t = ts + tt;
+    if (t >= L) break;
+    load_row8(xb + (long long)t * C, ring[(tt + W - 1) % W]);
+    const bf16x8 d = *(const bf16x8*)(dyb + (long long)t * C);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float z = bb[j];
+#pragma unroll
+      for (int wi = 0; wi < W; ++wi)
+        z += wf[j][wi] * ring[(tt + wi) % W][j];
+      const float s = 1.f / (1.f + __expf(-z));
+      o.v[j] = f2bf(bf2f(d.v[j]) * s * (1.f + z * (1.f - s)));
+    }
+    *(bf16x8*)(gb + (long long)t * C) = o;
   }
-  *(bf16x8*)(g + bt * C + c0) = o;
 }
 
-// pass 2: dx[t,c] = sum_i g[t + (W-1) - i, c] * w[c, i]
+// pass 2: dx[t,c] = sum_wi g[t + (W-1) - wi, c] * w[c, wi]
+// (future-looking window: rows t..t+W-1, slot of row r = (r - ts) % W)
 template <int W>
 __global__ void cconv_bwd_dx_kernel(const short* __restrict__ g,
                                     const short* __restrict__ w,
                                     short* __restrict__ dx,
-                                    int L, int C, long long total8) {
+                                    int L, int C, long long nstrip) {
   const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= total8) return;
-  const int C8 = C / 8;
-  const long long bt = i / C8;
-  const int c0 = (int)(i % C8) * 8;
-  const int t = (int)(bt % L);
-  const long long row0 = bt - t;
-  WBlock<W> wb;
-  wb.load(w, c0);
-  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  if (i >= nstrip) return;
+  STRIP_DECODE();
+  const short* gb = g + b_ * L * (long long)C + c0;
+  short* dxb = dx + b_ * L * (long long)C + c0;
+  float wf[8][W];
+  unpack_wb<W>(w, c0, wf);
+  float ring[W][8];
 #pragma unroll
-  for (int wi = 0; wi < W; ++wi) {
-    const int ti = t + (W - 1) - wi;
-    if (ti >= L) continue;
-    const bf16x8 gv = *(const bf16x8*)(g + (row0 + ti) * C + c0);
+  for (int k = 0; k < W - 1; ++k) {   // rows ts..ts+W-2
+    const int ti = ts + k;
+    if (ti < L)
+      load_row8(gb + (long long)ti * C, ring[k % W]);
+    else
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      acc[j] += bf2f(gv.v[j]) * wb.at(j, wi);
+      for (int j = 0; j < 8; ++j) ring[k % W][j] = 0.f;
   }
-  bf16x8 o;
 #pragma unroll
-  for (int j = 0; j < 8; ++j) o.v[j] = f2bf(acc[j]);
-  *(bf16x8*)(dx + bt * C + c0) = o;
+  for (int tt = 0; tt < 8; ++tt) {
+    const int t = ts + tt;
+    if (t >= L) break;
+    const int tin = t + W - 1;
+    if (tin < L)
+      load_row8(gb + (long long)tin * C, ring[(tt + W - 1) % W]);
+    else
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ring[(tt + W - 1) % W][j] = 0.f;
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float acc = 0.f;
+#pragma unroll
+      for (int wi = 0; wi < W; ++wi)   // row t+W-1-wi = slot (tt+W-1-wi)%W
+        acc += wf[j][wi] * ring[(tt + W - 1 - wi) % W][j];
+      o.v[j] = f2bf(acc);
+    }
+    *(bf16x8*)(dxb + (long long)t * C) = o;
+  }
 }
 
-// pass 3: dw[c,i] += sum_{b,t} g[t,c] x[t-W+1+i,c]; db[c] += sum g[t,c]
+// pass 3: dw[c,wi] += sum_{b,t} g[t,c] x[t-W+1+wi,c]; db[c] += sum g
+// 8-channel vector loads with the same sliding x window.
+template <int W>
 __global__ void cconv_bwd_dwdb_kernel(const short* __restrict__ g,
                                       const short* __restrict__ x,
                                       float* __restrict__ dw,
                                       float* __restrict__ db,
-                                      int L, int C, int W,
+                                      int L, int C,
                                       long long rows, int rows_per_chunk) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  const int c0 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c0 >= C) return;
   const long long r0 = (long long)blockIdx.y * rows_per_chunk;
   const long long r1 = min(r0 + rows_per_chunk, rows);
-  float accw[4] = {0.f, 0.f, 0.f, 0.f};  // W <= 4
-  float accb = 0.f;
+  float accw[W][8];
+#pragma unroll
+  for (int wi = 0; wi < W; ++wi)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) accw[wi][j] = 0.f;
+  float accb[8] = {0.f};
+  float ring[W][8];
+  int ring_t = -1 << 30;   // row index held in the newest slot
   for (long long bt = r0; bt < r1; ++bt) {
     const int t = (int)(bt % L);
     const long long row0 = bt - t;
-    const float gv = bf2f(g[bt * C + c]);
-    accb += gv;
+    // (re)fill the window when the sequence or chunk boundary breaks it
+    if (bt == r0 || t == 0 || ring_t != t - 1) {
+#pragma unroll
+      for (int k = 0; k < W - 1; ++k) {
+        const int ti = t - (W - 1) + k;
+        if (ti >= 0)
+          load_row8(x + (row0 + ti) * C + c0, ring[(ti % W + W) % W]);
+        else {
+          const int slot = ((ti % W) + W) % W;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) ring[slot][j] = 0.f;
+        }
+      }
+    }
+    load_row8(x + bt * C + c0, ring[t % W]);
+    ring_t = t;
+    float gv[8];
+    load_row8(g + bt * C + c0, gv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) accb[j] += gv[j];
+#pragma unroll
     for (int wi = 0; wi < W; ++wi) {
-      const int ti = t - W + 1 + wi;
-      if (ti >= 0) accw[wi] += gv * bf2f(x[(row0 + ti) * C + c]);
+      const int ti = t - (W - 1) + wi;
+      const int slot = ((ti % W) + W) % W;
+      if (ti >= 0)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) accw[wi][j] += gv[j] * ring[slot][j];
     }
   }
-  for (int wi = 0; wi < W; ++wi)
-    atomicAdd(&dw[(long long)c * W + wi], accw[wi]);
-  atomicAdd(&db[c], accb);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+#pragma unroll
+    for (int wi = 0; wi < W; ++wi)
+      atomicAdd(&dw[(long long)(c0 + j) * W + wi], accw[wi][j]);
+    atomicAdd(&db[c0 + j], accb[j]);
+  }
 }
 
 extern "C" {
@@ -171,50 +268,52 @@ extern "C" {
 void launch_cconv_fwd(const void* x, const void* w, const float* bias,
                       void* y, int BL, int L, int C, int W,
                       hipStream_t stream) {
-  const long long total8 = (long long)BL * (C / 8);
+  const int B = BL / L;
+  const long long nstrip = (long long)B * ((L + 7) / 8) * (C / 8);
   const int block = 256;
-  const int grid = (int)((total8 + block - 1) / block);
+  const int grid = (int)((nstrip + block - 1) / block);
   if (W == 4)
     cconv_fwd_kernel<4><<<grid, block, 0, stream>>>(
-        (const short*)x, (const short*)w, bias, (short*)y, L, C, total8);
+        (const short*)x, (const short*)w, bias, (short*)y, L, C, nstrip);
   else if (W == 3)
     cconv_fwd_kernel<3><<<grid, block, 0, stream>>>(
-        (const short*)x, (const short*)w, bias, (short*)y, L, C, total8);
+        (const short*)x, (const short*)w, bias, (short*)y, L, C, nstrip);
   else
     cconv_fwd_kernel<2><<<grid, block, 0, stream>>>(
-        (const short*)x, (const short*)w, bias, (short*)y, L, C, total8);
+        (const short*)x, (const short*)w, bias, (short*)y, L, C, nstrip);
 }
 
 void launch_cconv_bwd(const void* dy, const void* x, const void* w,
                       const float* bias, void* g, void* dx, float* dw,
                       float* db, int BL, int L, int C, int W,
                       hipStream_t stream) {
-  const long long total8 = (long long)BL * (C / 8);
+  const int B = BL / L;
+  const long long nstrip = (long long)B * ((L + 7) / 8) * (C / 8);
   const int block = 256;
-  const int grid = (int)((total8 + block - 1) / block);
-  if (W == 4) {
-    cconv_bwd_g_kernel<4><<<grid, block, 0, stream>>>(
-        (const short*)dy, (const short*)x, (const short*)w, bias, (short*)g,
-        L, C, total8);
-    cconv_bwd_dx_kernel<4><<<grid, block, 0, stream>>>(
-        (const short*)g, (const short*)w, (short*)dx, L, C, total8);
-  } else if (W == 3) {
-    cconv_bwd_g_kernel<3><<<grid, block, 0, stream>>>(
-        (const short*)dy, (const short*)x, (const short*)w, bias, (short*)g,
-        L, C, total8);
-    cconv_bwd_dx_kernel<3><<<grid, block, 0, stream>>>(
-        (const short*)g, (const short*)w, (short*)dx, L, C, total8);
-  } else {
-    cconv_bwd_g_kernel<2><<<grid, block, 0, stream>>>(
-        (const short*)dy, (const short*)x, (const short*)w, bias, (short*)g,
-        L, C, total8);
-    cconv_bwd_dx_kernel<2><<<grid, block, 0, stream>>>(
-        (const short*)g, (const short*)w, (short*)dx, L, C, total8);
-  }
+  const int grid = (int)((nstrip + block - 1) / block);
+#define CCONV_BWD(WV)                                                       \
+  do {                                                                      \
+    cconv_bwd_g_kernel<WV><<<grid, block, 0, stream>>>(                     \
+        (const short*)dy, (const short*)x, (const short*)w, bias,           \
+        (short*)g, L, C, nstrip);                                           \
+    cconv_bwd_dx_kernel<WV><<<grid, block, 0, stream>>>(                    \
+        (const short*)g, (const short*)w, (short*)dx, L, C, nstrip);        \
+  } while (0)
+  if (W == 4) CCONV_BWD(4);
+  else if (W == 3) CCONV_BWD(3);
+  else CCONV_BWD(2);
+#undef CCONV_BWD
   const int rpc = max(1, (int)((BL + 63) / 64));
-  dim3 g2((C + 255) / 256, (BL + rpc - 1) / rpc);
-  cconv_bwd_dwdb_kernel<<<g2, 256, 0, stream>>>(
-      (const short*)g, (const short*)x, dw, db, L, C, W, BL, rpc);
+  dim3 g2((C / 8 + 255) / 256, (BL + rpc - 1) / rpc);
+  if (W == 4)
+    cconv_bwd_dwdb_kernel<4><<<g2, 256, 0, stream>>>(
+        (const short*)g, (const short*)x, dw, db, L, C, BL, rpc);
+  else if (W == 3)
+    cconv_bwd_dwdb_kernel<3><<<g2, 256, 0, stream>>>(
+        (const short*)g, (const short*)x, dw, db, L, C, BL, rpc);
+  else
+    cconv_bwd_dwdb_kernel<2><<<g2, 256, 0, stream>>>(
+        (const short*)g, (const short*)x, dw, db, L, C, BL, rpc);
 }
 
 }  // extern "C"
