@@ -216,30 +216,34 @@ class Mamba2Mixer(nn.Module):
         x2 = x.reshape(b * l, h * p)
         z2 = z.reshape(b * l, h * p)
         dtf, dacs = ops.ssd_prep(dt2, dtb, Alog, Q)        # (b*nc*h, Q) fp32
+        # xdt/xdtd come out H-MAJOR (b,nc,h,Q,p): y_diag is then a plain
+        # strided-batched bmm and no einsum materializes permuted copies
         xdt, xdtd = ops.ssd_xdt(x2, dtf, dacs, h, p, Q)    # bf16 pair
+        N = b * nc * h
         Bm = B.reshape(b, nc, Q, g, n)
         Cm = C.reshape(b, nc, Q, g, n)
         # scores per GROUP (g << h), decay folded in-kernel (L never
         # materialized; d_scores comes back from the fused backward)
         scores = torch.einsum("bcqgn,bckgn->bcgqk", Cm, Bm).contiguous()
         sL = ops.ssd_scores_decay(dacs, scores.view(b * nc * g, Q, Q), h, g)
-        y_diag = torch.einsum(
-            "bcgrqk,bckgrp->bcqgrp", sL.view(b, nc, g, rep, Q, Q),
-            xdt.view(b, nc, Q, g, rep, p))                 # bf16
+        y_diag = torch.bmm(sL, xdt.view(N, Q, p))          # (N, Q, p) bf16
         states = torch.einsum(
-            "bckgn,bckgrp->bcgrnp", Bm,
-            xdtd.view(b, nc, Q, g, rep, p)).float().reshape(b, nc, h, n, p)
+            "bckgn,bcgrkp->bcgrnp", Bm,
+            xdtd.view(b, nc, g, rep, Q, p)).float().reshape(b, nc, h, n, p)
         # inter-chunk recurrence in closed form (see ssd_chunked)
         G = dacs.view(b, nc, h, Q)[..., -1].permute(0, 2, 1)
         W = torch.exp(segsum(G))
         P_ = torch.einsum("bhzc,bchnp->bzhnp", W, states)
         prev = torch.cat([torch.zeros_like(P_[:, :1]), P_[:, :-1]], dim=1)
+        # output ordering "bcgqrp" = the bmm-natural layout; the ygate
+        # kernel reads it in place
         y_off = torch.einsum(
-            "bcqgn,bcgrnp->bcqgrp", Cm,
+            "bcqgn,bcgrnp->bcgqrp", Cm,
             prev.view(b, nc, g, rep, n, p).to(torch.bfloat16))
         out = ops.ssd_ygate(
-            y_diag.reshape(b * l, h * p), y_off.reshape(b * l, h * p),
-            dacs, x2, D_, z2, h, p, Q)
+            y_diag.reshape(b * l, h * p),
+            y_off.contiguous().view(b * l, h * p),
+            dacs, x2, D_, z2, h, g, p, Q)
         return out.view(b, l, h * p)
 
     def forward(self, u):

@@ -527,7 +527,8 @@ def ssd_prep(dt2d, dt_bias, A_log, chunk):
 
 class _SSDXdtFn(torch.autograd.Function):
     """x2d (b*l, H*P) strided bf16 -> (xdt, xdt_decayed) bf16 contiguous
-    (layout (b,nc,Q,H,P)); xdt = x*dtf, decayed by exp(dacs_end-dacs)."""
+    in H-MAJOR layout (b,nc,H,Q,P) — bmm-native for y_diag (no einsum
+    permute copies); xdt = x*dtf, decayed by exp(dacs_end-dacs)."""
 
     @staticmethod
     def forward(ctx, x2d, dtf, dacs, H, P, Q):
@@ -589,26 +590,29 @@ class _SSDYGateFn(torch.autograd.Function):
     (y assembly + D residual + the gated epilogue)."""
 
     @staticmethod
-    def forward(ctx, ydiag, yoff, dacs, x2d, D, z2d, H, P, Q):
+    def forward(ctx, ydiag, yoff, dacs, x2d, D, z2d, H, G, P, Q):
         ext = _require_ext("ssd_ygate")
         D_f = D.detach().float().contiguous()
-        out = ext.ssd_ygate_fwd(ydiag, yoff, dacs, x2d, D_f, z2d, H, P, Q)
+        out = ext.ssd_ygate_fwd(ydiag, yoff, dacs, x2d, D_f, z2d, H, G, P, Q)
         ctx.save_for_backward(ydiag, yoff, dacs, x2d, D_f, z2d)
-        ctx.meta = (H, P, Q, D.dtype)
+        ctx.meta = (H, G, P, Q, D.dtype)
         return out
 
     @staticmethod
     def backward(ctx, dout):
         ydiag, yoff, dacs, x2d, D_f, z2d = ctx.saved_tensors
-        H, P, Q, ddt = ctx.meta
+        H, G, P, Q, ddt = ctx.meta
         dydiag, dyoff, ddacs, dx, dD_rows, dz = _C.ssd_ygate_bwd(
-            dout.contiguous(), ydiag, yoff, dacs, x2d, D_f, z2d, H, P, Q)
+            dout.contiguous(), ydiag, yoff, dacs, x2d, D_f, z2d, H, G, P, Q)
         dD = dD_rows.sum(0)
-        return (dydiag, dyoff, ddacs, dx, dD.to(ddt), dz, None, None, None)
+        return (dydiag, dyoff, ddacs, dx, dD.to(ddt), dz, None, None, None,
+                None)
 
 
-def ssd_ygate(ydiag, yoff, dacs, x2d, D, z2d, H, P, Q):
-    return _SSDYGateFn.apply(ydiag, yoff, dacs, x2d, D, z2d, H, P, Q)
+def ssd_ygate(ydiag, yoff, dacs, x2d, D, z2d, H, G, P, Q):
+    """ydiag h-major (b,nc,h,Q,p) flat; yoff in the y_off einsum's
+    natural (b,nc,g,Q,rep,p) layout; out (b, l, H*P) row-major."""
+    return _SSDYGateFn.apply(ydiag, yoff, dacs, x2d, D, z2d, H, G, P, Q)
 
 
 # --------------------------------------------------------------------------

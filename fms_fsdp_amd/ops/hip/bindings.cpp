@@ -58,11 +58,11 @@ void launch_ssd_sl_bwd(const void*, const void*, const float*, void*, float*,
                        long long, int, int, int, hipStream_t);
 void launch_ssd_ygate_fwd(const void*, const void*, const float*, const void*,
                           const float*, const void*, void*, long long, int,
-                          int, int, long long, long long, hipStream_t);
+                          int, int, int, long long, long long, hipStream_t);
 void launch_ssd_ygate_bwd(const void*, const void*, const void*, const float*,
                           const void*, const float*, const void*, void*,
                           void*, float*, void*, float*, void*, long long, int,
-                          int, int, long long, long long, long long,
+                          int, int, int, long long, long long, long long,
                           long long, hipStream_t);
 }
 
@@ -439,7 +439,8 @@ std::tuple<Tensor, Tensor> ssd_sl_bwd(Tensor g, Tensor scores, Tensor dacs,
 }
 
 Tensor ssd_ygate_fwd(Tensor ydiag, Tensor yoff, Tensor dacs, Tensor x,
-                     Tensor Dp, Tensor z, int64_t H, int64_t P, int64_t Q) {
+                     Tensor Dp, Tensor z, int64_t H, int64_t G, int64_t P,
+                     int64_t Q) {
   const long long sx = slice_stride(x, "x");
   const long long sz = slice_stride(z, "z");
   TORCH_CHECK(ydiag.is_contiguous() && yoff.is_contiguous());
@@ -448,14 +449,15 @@ Tensor ssd_ygate_fwd(Tensor ydiag, Tensor yoff, Tensor dacs, Tensor x,
   launch_ssd_ygate_fwd(ydiag.data_ptr(), yoff.data_ptr(),
                        dacs.data_ptr<float>(), x.data_ptr(),
                        Dp.data_ptr<float>(), z.data_ptr(), out.data_ptr(),
-                       rows * H * P / 8, (int)H, (int)Q, (int)P, sx, sz,
-                       cur_stream());
+                       rows * H * P / 8, (int)H, (int)G, (int)Q, (int)P, sx,
+                       sz, cur_stream());
   return out;
 }
 
 std::vector<Tensor> ssd_ygate_bwd(Tensor dout, Tensor ydiag, Tensor yoff,
                                   Tensor dacs, Tensor x, Tensor Dp, Tensor z,
-                                  int64_t H, int64_t P, int64_t Q) {
+                                  int64_t H, int64_t G, int64_t P,
+                                  int64_t Q) {
   const long long sx = slice_stride(x, "x");
   const long long sz = slice_stride(z, "z");
   const long long rows = x.size(0);
@@ -471,8 +473,8 @@ std::vector<Tensor> ssd_ygate_bwd(Tensor dout, Tensor ydiag, Tensor yoff,
                        Dp.data_ptr<float>(), z.data_ptr(), dydiag.data_ptr(),
                        dyoff.data_ptr(), ddacs.data_ptr<float>(),
                        dx.data_ptr(), dD_rows.data_ptr<float>(),
-                       dz.data_ptr(), rows * H, (int)H, (int)Q, (int)P, sx,
-                       sz, H * P, H * P, cur_stream());
+                       dz.data_ptr(), rows * H, (int)H, (int)G, (int)Q,
+                       (int)P, sx, sz, H * P, H * P, cur_stream());
   return {dydiag, dyoff, ddacs, dx, dD_rows, dz};
 }
 

@@ -195,8 +195,11 @@ __global__ void ssd_prep_bwd_kernel(const float* __restrict__ ddtf,
 }
 
 // ---------------------------------------------------------------------
-// ssd_xdt: xdt[b,c,q,h,p]     = x[b, cQ+q, h*P+p] * dtf[n,q]   (bf16)
-//          xdtd[b,c,q,h,p]    = xdt * exp(dacs[n,Q-1] - dacs[n,q])
+// ssd_xdt: xdt[b,c,h,q,p]     = x[b, cQ+q, h*P+p] * dtf[n,q]   (bf16)
+//          xdtd[b,c,h,q,p]    = xdt * exp(dacs[n,Q-1] - dacs[n,q])
+// H-MAJOR layout: (b,nc,h,Q,p) is exactly torch.bmm's batch layout for
+// y_diag = sL (N,Q,Q) @ xdt (N,Q,p) — the previous q-major layout made
+// every downstream einsum materialize a permuted copy (~64 MB each).
 // vectorized 8-wide over p (P % 8 == 0).
 // ---------------------------------------------------------------------
 __global__ void ssd_xdt_fwd_kernel(const short* __restrict__ x,
@@ -210,11 +213,11 @@ __global__ void ssd_xdt_fwd_kernel(const short* __restrict__ x,
   if (idx >= total8) return;
   const int P8 = P / 8;
   const int p0 = (int)(idx % P8) * 8;
-  long long r = idx / P8;                  // (bc, q, h)
-  const int h = (int)(r % H);
-  r /= H;
+  long long r = idx / P8;                  // (bc, h, q)
   const int q = (int)(r % Q);
-  const long long bc = r / Q;
+  r /= Q;
+  const int h = (int)(r % H);
+  const long long bc = r / H;
   const long long n = bc * H + h;
   const float f = dtf[n * Q + q];
   const float dec = __expf(dacs[n * Q + Q - 1] - dacs[n * Q + q]);
@@ -226,7 +229,7 @@ __global__ void ssd_xdt_fwd_kernel(const short* __restrict__ x,
     o1.v[e] = f2bf(xf);
     o2.v[e] = f2bf(xf * dec);
   }
-  const long long out = ((bc * Q + q) * H + h) * (long long)P + p0;
+  const long long out = (n * Q + q) * (long long)P + p0;
   *(bf16x8*)(xdt + out) = o1;
   *(bf16x8*)(xdtd + out) = o2;
 }
@@ -250,14 +253,14 @@ __global__ void ssd_xdt_bwd_kernel(const short* __restrict__ dxdt,
   const long long r = ((long long)blockIdx.x * blockDim.x + threadIdx.x) / 64;
   if (r >= nrows) return;
   const int lane = threadIdx.x & 63;
-  const int h = (int)(r % H);
-  const long long bcq = r / H;
-  const int q = (int)(bcq % Q);
-  const long long bc = bcq / Q;
-  const long long n = bc * H + h;
+  const int q = (int)(r % Q);
+  const long long bch = r / Q;
+  const int h = (int)(bch % H);
+  const long long bc = bch / H;
+  const long long n = bch;
   const float f = dtf[n * Q + q];
   const float dec = __expf(dacs[n * Q + Q - 1] - dacs[n * Q + q]);
-  const long long go = ((bc * Q + q) * H + h) * (long long)P;
+  const long long go = (n * Q + q) * (long long)P;   // h-major grads
   const long long xo = (bc * Q + q) * sx + h * P;
   const long long dxo = (bc * Q + q) * sdx + h * P;
   float sdtf = 0.f, sdec = 0.f;
@@ -381,6 +384,10 @@ __global__ void ssd_sl_bwd_col_kernel(const short* __restrict__ g,
 // ssd_ygate: y = ydiag + yoff*exp(dacs) + x*D;  out = y * silu(z)
 // x and z are strided bf16 slices; out is (b, l, H*P) bf16 contiguous.
 // ---------------------------------------------------------------------
+// ydiag arrives H-MAJOR (b,nc,h,Q,p) (bmm-native), yoff arrives in the
+// y_off einsum's natural output layout (b,nc,g,Q,rep,p) — reading each
+// operand in its producer's layout removes the permute copies torch
+// would otherwise materialize. out stays (b, l, H*P) row-major.
 __global__ void ssd_ygate_fwd_kernel(const short* __restrict__ ydiag,
                                      const short* __restrict__ yoff,
                                      const float* __restrict__ dacs,
@@ -388,8 +395,9 @@ __global__ void ssd_ygate_fwd_kernel(const short* __restrict__ ydiag,
                                      const float* __restrict__ Dp,
                                      const short* __restrict__ z,
                                      short* __restrict__ out,
-                                     int H, int Q, int P, long long sx,
-                                     long long sz, long long total8) {
+                                     int H, int G, int Q, int P,
+                                     long long sx, long long sz,
+                                     long long total8) {
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= total8) return;
   const int P8 = P / 8;
@@ -400,11 +408,14 @@ __global__ void ssd_ygate_fwd_kernel(const short* __restrict__ ydiag,
   const int q = (int)(r % Q);
   const long long bc = r / Q;
   const long long n = bc * H + h;
+  const int rep = H / G;
   const float sd = __expf(dacs[n * Q + q]);
   const float Dh = Dp[h];
-  const long long go = ((bc * Q + q) * H + h) * (long long)P + p0;
-  const bf16x8 yd = *(const bf16x8*)(ydiag + go);
-  const bf16x8 yo = *(const bf16x8*)(yoff + go);
+  const long long yd_o = (n * Q + q) * (long long)P + p0;
+  const long long yo_o =
+      ((((bc * G + h / rep) * Q + q) * rep + h % rep)) * (long long)P + p0;
+  const bf16x8 yd = *(const bf16x8*)(ydiag + yd_o);
+  const bf16x8 yo = *(const bf16x8*)(yoff + yo_o);
   const bf16x8 xv = *(const bf16x8*)(x + (bc * Q + q) * sx + h * P + p0);
   const bf16x8 zv = *(const bf16x8*)(z + (bc * Q + q) * sz + h * P + p0);
   bf16x8 o;
@@ -414,7 +425,7 @@ __global__ void ssd_ygate_fwd_kernel(const short* __restrict__ ydiag,
     const float zf = bf2f(zv.v[e]);
     o.v[e] = f2bf(y * zf * sigmoidf(zf));
   }
-  *(bf16x8*)(out + go) = o;
+  *(bf16x8*)(out + ((bc * Q + q) * (long long)H + h) * P + p0) = o;
 }
 
 // backward: one wave per (b,c,q,h) row, reduction over p. Each row owns
@@ -435,9 +446,10 @@ __global__ void ssd_ygate_bwd_kernel(const short* __restrict__ dout,
                                      short* __restrict__ dx,
                                      float* __restrict__ dD_rows,
                                      short* __restrict__ dz,
-                                     int H, int Q, int P, long long sx,
-                                     long long sz, long long sdx,
-                                     long long sdz, long long nrows) {
+                                     int H, int G, int Q, int P,
+                                     long long sx, long long sz,
+                                     long long sdx, long long sdz,
+                                     long long nrows) {
   const long long r = ((long long)blockIdx.x * blockDim.x + threadIdx.x) / 64;
   if (r >= nrows) return;
   const int lane = threadIdx.x & 63;
@@ -446,9 +458,13 @@ __global__ void ssd_ygate_bwd_kernel(const short* __restrict__ dout,
   const int q = (int)(bcq % Q);
   const long long bc = bcq / Q;
   const long long n = bc * H + h;
+  const int rep = H / G;
   const float sd = __expf(dacs[n * Q + q]);
   const float Dh = Dp[h];
-  const long long go = ((bc * Q + q) * H + h) * (long long)P;
+  const long long go = ((bc * Q + q) * H + h) * (long long)P;   // out layout
+  const long long yd_o = (n * Q + q) * (long long)P;            // h-major
+  const long long yo_o =
+      (((bc * G + h / rep) * Q + q) * rep + h % rep) * (long long)P;
   const long long xo = (bc * Q + q) * sx + h * P;
   const long long zo = (bc * Q + q) * sz + h * P;
   const long long dxo = (bc * Q + q) * sdx + h * P;
@@ -459,12 +475,12 @@ __global__ void ssd_ygate_bwd_kernel(const short* __restrict__ dout,
     const float zf = bf2f(z[zo + p]);
     const float sg = sigmoidf(zf);
     const float sil = zf * sg;
-    const float yofp = bf2f(yoff[go + p]);
+    const float yofp = bf2f(yoff[yo_o + p]);
     const float xf = bf2f(x[xo + p]);
-    const float y = bf2f(ydiag[go + p]) + yofp * sd + xf * Dh;
+    const float y = bf2f(ydiag[yd_o + p]) + yofp * sd + xf * Dh;
     const float dy = gy * sil;
-    dydiag[go + p] = f2bf(dy);
-    dyoff[go + p] = f2bf(dy * sd);
+    dydiag[yd_o + p] = f2bf(dy);
+    dyoff[yo_o + p] = f2bf(dy * sd);
     dx[dxo + p] = f2bf(dy * Dh);
     dz[dzo + p] = f2bf(gy * y * sg * (1.f + zf * (1.f - sg)));
     sdd += dy * yofp * sd;
@@ -476,6 +492,53 @@ __global__ void ssd_ygate_bwd_kernel(const short* __restrict__ dout,
     ddacs[n * Q + q] = sdd;
     dD_rows[r] = sD;
   }
+}
+
+// ---------------------------------------------------------------------
+// Fused sL backward for Q=128: one block per n computes, in a single
+// read of g, (a) the per-head d_scores = g*L (bf16 out), and (b)
+// d_cs[i] = rowsum(g*s*L)[i] - colsum(g*s*L)[i] via an LDS-staged
+// product tile. Replaces the 3-pass version whose column pass read g
+// with a Q-stride (412 us/call, the single largest mamba bwd kernel
+// after the gate fix).
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void ssd_sl_bwd_fused_kernel(
+    const short* __restrict__ g, const short* __restrict__ scores,
+    const float* __restrict__ cs, short* __restrict__ dsh,
+    float* __restrict__ dcs, int H, int G) {
+  constexpr int Q = 128;
+  __shared__ float prod[Q][Q + 5];   // g*s*L; +5 pad: row-sum lane stride 133 is coprime to the 64 banks
+  __shared__ float csl[Q];
+  __shared__ float rsum[Q], csum[Q];
+  const long long n = blockIdx.x;
+  const long long m = (n / H) * G + (int)(n % H) / (H / G);
+  const int tid = threadIdx.x;
+  if (tid < Q) csl[tid] = cs[n * Q + tid];
+  __syncthreads();
+  // stage: 256 threads x 64 elements, consecutive j per iteration
+#pragma unroll
+  for (int k = 0; k < Q * Q / 256; ++k) {
+    const int idx = k * 256 + tid;
+    const int i = idx >> 7;
+    const int j = idx & (Q - 1);
+    const float gv = bf2f(g[(n * Q + i) * Q + j]);
+    const float L = (j <= i) ? __expf(csl[i] - csl[j]) : 0.f;
+    dsh[(n * Q + i) * Q + j] = f2bf(gv * L);
+    prod[i][j] = gv * L * bf2f(scores[(m * Q + i) * Q + j]);
+  }
+  __syncthreads();
+  if (tid < Q) {           // row sums
+    float a = 0.f;
+    for (int j = 0; j <= tid; ++j) a += prod[tid][j];
+    rsum[tid] = a;
+  } else {                 // column sums (padded rows rotate banks)
+    const int j = tid - Q;
+    float a = 0.f;
+    for (int i = j; i < Q; ++i) a += prod[i][j];
+    csum[j] = a;
+  }
+  __syncthreads();
+  if (tid < Q) dcs[n * Q + tid] += rsum[tid] - csum[tid];
 }
 
 extern "C" {
@@ -554,6 +617,11 @@ void launch_ssd_sl_bwd(const void* g, const void* scores, const float* cs,
                        void* dsh, float* dcs, long long N, int H, int G,
                        int Q, hipStream_t stream) {
   const int block = 256;
+  if (Q == 128) {
+    ssd_sl_bwd_fused_kernel<<<(int)N, block, 0, stream>>>(
+        (const short*)g, (const short*)scores, cs, (short*)dsh, dcs, H, G);
+    return;
+  }
   const long long total8 = N * Q * (Q / 8);
   ssd_sl_bwd_dscores_kernel<<<(int)((total8 + block - 1) / block), block, 0,
                               stream>>>((const short*)g, cs, (short*)dsh, Q,
@@ -569,29 +637,30 @@ void launch_ssd_sl_bwd(const void* g, const void* scores, const float* cs,
 void launch_ssd_ygate_fwd(const void* ydiag, const void* yoff,
                           const float* dacs, const void* x, const float* Dp,
                           const void* z, void* out, long long total8, int H,
-                          int Q, int P, long long sx, long long sz,
+                          int G, int Q, int P, long long sx, long long sz,
                           hipStream_t stream) {
   const int block = 256;
   ssd_ygate_fwd_kernel<<<(int)((total8 + block - 1) / block), block, 0,
                          stream>>>((const short*)ydiag, (const short*)yoff,
                                    dacs, (const short*)x, Dp,
-                                   (const short*)z, (short*)out, H, Q, P, sx,
-                                   sz, total8);
+                                   (const short*)z, (short*)out, H, G, Q, P,
+                                   sx, sz, total8);
 }
 
 void launch_ssd_ygate_bwd(const void* dout, const void* ydiag,
                           const void* yoff, const float* dacs, const void* x,
                           const float* Dp, const void* z, void* dydiag,
                           void* dyoff, float* ddacs, void* dx, float* dD,
-                          void* dz, long long nrows, int H, int Q, int P,
-                          long long sx, long long sz, long long sdx,
+                          void* dz, long long nrows, int H, int G, int Q,
+                          int P, long long sx, long long sz, long long sdx,
                           long long sdz, hipStream_t stream) {
   const int block = 256;
   const int grid = (int)((nrows * 64 + block - 1) / block);
   ssd_ygate_bwd_kernel<<<grid, block, 0, stream>>>(
       (const short*)dout, (const short*)ydiag, (const short*)yoff, dacs,
       (const short*)x, Dp, (const short*)z, (short*)dydiag, (short*)dyoff,
-      ddacs, (short*)dx, dD, (short*)dz, H, Q, P, sx, sz, sdx, sdz, nrows);
+      ddacs, (short*)dx, dD, (short*)dz, H, G, Q, P, sx, sz, sdx, sdz,
+      nrows);
 }
 
 }  // extern "C"
