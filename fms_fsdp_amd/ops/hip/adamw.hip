@@ -27,8 +27,10 @@ __global__ void adamw_kernel(float* __restrict__ p,
                              long long n4, float lr, float b1, float b2,
                              float eps, float wd, float bc1, float bc2,
                              const float* __restrict__ gscale) {
-  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n4) return;
+  // grid-stride: ~8 quads per thread on the pooled 7B shard keeps the
+  // stream engines fed without a 1.7M-block launch
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n4; i += (long long)gridDim.x * blockDim.x) {
   const float gs = gscale ? *gscale : 1.f;
   // every stream here is touched exactly once per step (189 GB/step on
   // a 7B shard): nontemporal hints keep them from thrashing L2
@@ -68,6 +70,7 @@ __global__ void adamw_kernel(float* __restrict__ p,
 #pragma unroll
     for (int j = 0; j < 4; ++j) o[j] = (ODT == 1) ? f2bf(pv[j]) : f2h(pv[j]);
     __builtin_nontemporal_store(o, (s16x4v*)p_lowp_out + i);
+  }
   }
 }
 
