@@ -498,9 +498,18 @@ def test_checkpoint_reload_match_persistent_workers(corpus, tmp_path):
     for _ in range(100):
         for ld in loaders:
             next(ld)
-    ckps = os.listdir(os.path.join(ck_root, "checkpoints"))
-    assert len(ckps) == 1, ckps
-    shards = os.listdir(os.path.join(ck_root, "checkpoints", ckps[0]))
+    # worker processes write the checkpoint asynchronously relative to
+    # the main process's consumption (prefetch): wait for all 3 shards
+    import time
+    deadline = time.time() + 30
+    shards = []
+    while time.time() < deadline:
+        ckps = os.listdir(os.path.join(ck_root, "checkpoints"))
+        if len(ckps) == 1:
+            shards = os.listdir(os.path.join(ck_root, "checkpoints", ckps[0]))
+            if len(shards) == 3:
+                break
+        time.sleep(0.2)
     assert len(shards) == 3, shards
 
     ds2 = mk(1000)
