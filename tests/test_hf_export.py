@@ -57,7 +57,15 @@ def test_dcp_import_roundtrip(tmp_path):
     ref = Llama(cfg)
     ref.reset_parameters()
 
-    # synthesize the reference's DCP layout: fms names, unfused q/k/v
+    def half_to_interleave(w, n_heads):
+        # inverse of the importer's row re-order: our half-rotation
+        # layout back to fms's interleaved pairs
+        rows, cols = w.shape
+        return w.view(n_heads, 2, rows // n_heads // 2, cols) \
+            .transpose(1, 2).reshape(rows, cols)
+
+    # synthesize the reference's DCP layout: fms names, unfused q/k/v,
+    # INTERLEAVED RoPE rows (what a real fms checkpoint holds)
     hd = cfg.nheads * cfg.head_dim
     fms_sd = {"base_model.embedding.weight": ref.embedding.weight.detach(),
               "head.weight": ref.lm_head.weight.detach(),
@@ -67,8 +75,10 @@ def test_dcp_import_roundtrip(tmp_path):
         qkv = layer.attn.qkv.weight.detach()
         fms_sd[p + "ln.weight"] = layer.attn_norm.weight.detach()
         fms_sd[p + "ff_ln.weight"] = layer.mlp_norm.weight.detach()
-        fms_sd[p + "attn.in_proj.query.weight"] = qkv[:hd]
-        fms_sd[p + "attn.in_proj.key.weight"] = qkv[hd:2 * hd]
+        fms_sd[p + "attn.in_proj.query.weight"] = \
+            half_to_interleave(qkv[:hd], cfg.nheads)
+        fms_sd[p + "attn.in_proj.key.weight"] = \
+            half_to_interleave(qkv[hd:2 * hd], cfg.kvheads)
         fms_sd[p + "attn.in_proj.value.weight"] = qkv[2 * hd:]
         fms_sd[p + "attn.dense.weight"] = layer.attn.proj.weight.detach()
         wg1 = layer.mlp.wg1.weight.detach()
@@ -79,7 +89,8 @@ def test_dcp_import_roundtrip(tmp_path):
     dcp.save({"model_state": fms_sd}, checkpoint_id=dcp_dir, no_dist=True)
 
     out = str(tmp_path / "imported.pth")
-    mapped = conv.convert(dcp_dir, out)
+    mapped = conv.convert(dcp_dir, out, nheads=cfg.nheads,
+                          kvheads=cfg.kvheads)
     assert set(mapped) == {n for n, _ in ref.named_parameters()}
 
     # load through the Checkpointer's single-file path into a new model

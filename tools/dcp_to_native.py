@@ -12,10 +12,11 @@ ibm-fms LLaMA state dict (reference: fms_fsdp/utils/checkpointing_utils.py
   3. writes a plain torch.save file that `Checkpointer.load(path=<file>)`
      / `main_training_llama --ckpt_load_path=<file>` accepts.
 
-No weight-layout permutation is applied: this framework shares the
-reference's interleaved q/k RoPE row convention (which is why
-fms_to_hf_llama.py here performs the same q/k permutation the reference's
-exporter does when going to HF).
+The q/k projection rows are re-ordered from ibm-fms's INTERLEAVED RoPE
+pair layout to this framework's half-rotation (HF-native) layout — the
+same `view(nh, d/2, 2, in).transpose(1, 2)` transform the reference's own
+HF exporter applies (fms_to_hf_llama.py:104-124 there). Pass
+--model_variant so the head geometry is known.
 
 Optimizer state is intentionally NOT imported: the reference's DCP
 optimizer payload is torch-FSDP flat-parameter sharded and meaningless to
@@ -39,8 +40,16 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import torch
 
 
-def map_fms_to_native(sd):
-    """fms LLaMA state dict -> our Llama names. Returns (mapped, skipped)."""
+def _interleave_to_half(w, n_heads):
+    """fms interleaved RoPE rows -> half-rotation rows (per head)."""
+    rows, cols = w.shape
+    return w.view(n_heads, rows // n_heads // 2, 2, cols) \
+        .transpose(1, 2).reshape(rows, cols)
+
+
+def map_fms_to_native(sd, nheads=None, kvheads=None):
+    """fms LLaMA state dict -> our Llama names (+ RoPE row re-order when
+    the head geometry is given). Returns (mapped, skipped)."""
     out = {}
     skipped = []
     qkv_parts = {}   # layer -> {q,k,v}
@@ -90,6 +99,18 @@ def map_fms_to_native(sd):
             f"layer {li}: incomplete q/k/v triple {list(parts)}"
         out[f"layers.{li}.attn.qkv.weight"] = torch.cat(
             [parts["q"], parts["k"], parts["v"]], dim=0)
+
+    if nheads is not None:
+        kvh = kvheads or nheads
+        for k in list(out):
+            if k.endswith("attn.qkv.weight"):
+                w = out[k]
+                qr = w.shape[0] * nheads // (nheads + 2 * kvh)
+                kr = (w.shape[0] - qr) // 2
+                out[k] = torch.cat(
+                    [_interleave_to_half(w[:qr], nheads),
+                     _interleave_to_half(w[qr:qr + kr], kvh),
+                     w[qr + kr:]], dim=0)
     for li, parts in gate_parts.items():
         assert set(parts) == {"wg", "w1"}, \
             f"layer {li}: incomplete wg/w1 pair {list(parts)}"
@@ -98,7 +119,7 @@ def map_fms_to_native(sd):
     return out, skipped
 
 
-def convert(dcp_path, out_path):
+def convert(dcp_path, out_path, nheads=None, kvheads=None):
     from torch.distributed.checkpoint.format_utils import dcp_to_torch_save
     with tempfile.TemporaryDirectory() as td:
         tmp = os.path.join(td, "flat.pth")
@@ -107,7 +128,10 @@ def convert(dcp_path, out_path):
     sd = raw.get("model_state", raw)
     if "_orig_mod" in sd:   # compiled-model nesting (fms_to_hf_llama.py:155)
         sd = sd["_orig_mod"]
-    mapped, skipped = map_fms_to_native(sd)
+    mapped, skipped = map_fms_to_native(sd, nheads, kvheads)
+    if nheads is None:
+        print("WARNING: no --model_variant given — q/k rows were NOT "
+              "re-ordered from the fms interleaved RoPE layout")
     if skipped:
         print(f"WARNING: {len(skipped)} unmapped keys (first 10): "
               f"{skipped[:10]}")
@@ -121,5 +145,12 @@ if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--dcp_path", required=True)
     ap.add_argument("--out", required=True)
+    ap.add_argument("--model_variant", default=None,
+                    help="registry name (e.g. llama2_7b) for head geometry")
     a = ap.parse_args()
-    convert(a.dcp_path, a.out)
+    nh = kvh = None
+    if a.model_variant:
+        from fms_fsdp_amd.config import get_model_config
+        mc = get_model_config(a.model_variant)
+        nh, kvh = mc.nheads, mc.kvheads
+    convert(a.dcp_path, a.out, nh, kvh)
