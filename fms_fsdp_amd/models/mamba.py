@@ -138,9 +138,9 @@ def ssd_chunked(x, dt, A, B, C, chunk):
     # Python loop (which cost thousands of tiny kernel launches).
     G = dA_cs[..., -1].permute(0, 2, 1)                         # (b,h,nc)
     W = torch.exp(segsum(G))                                    # (b,h,nc,nc)
-    P = torch.einsum("bhzc,bchnp->bzhnp", W, states)            # (b,nc,h,n,p)
-    prev_states = torch.cat(
-        [torch.zeros_like(P[:, :1]), P[:, :-1]], dim=1)         # (b,nc,h,n,p)
+    # shift the tiny decay matrix, not the big states tensor
+    Wsh = torch.cat([torch.zeros_like(W[:, :, :1]), W[:, :, :-1]], dim=2)
+    prev_states = torch.einsum("bhzc,bchnp->bzhnp", Wsh, states)  # (b,nc,h,n,p)
 
     # off-diagonal: Y_off[i] = C_i exp(dA_cs[i]) S_{c-1} — C stays per
     # group; the per-(q,h) decay multiplies the GEMM OUTPUT (size h*p)
@@ -230,11 +230,13 @@ class Mamba2Mixer(nn.Module):
         states = torch.einsum(
             "bckgn,bcgrkp->bcgrnp", Bm,
             xdtd.view(b, nc, g, rep, Q, p)).float().reshape(b, nc, h, n, p)
-        # inter-chunk recurrence in closed form (see ssd_chunked)
+        # inter-chunk recurrence in closed form (see ssd_chunked). The
+        # one-chunk shift lives on the TINY (b,h,nc,nc) decay matrix
+        # (zero first row) instead of a 167 MB cat on the states tensor.
         G = dacs.view(b, nc, h, Q)[..., -1].permute(0, 2, 1)
         W = torch.exp(segsum(G))
-        P_ = torch.einsum("bhzc,bchnp->bzhnp", W, states)
-        prev = torch.cat([torch.zeros_like(P_[:, :1]), P_[:, :-1]], dim=1)
+        Wsh = torch.cat([torch.zeros_like(W[:, :, :1]), W[:, :, :-1]], dim=2)
+        prev = torch.einsum("bhzc,bchnp->bzhnp", Wsh, states)
         # output ordering "bcgqrp" = the bmm-natural layout; the ygate
         # kernel reads it in place
         y_off = torch.einsum(
