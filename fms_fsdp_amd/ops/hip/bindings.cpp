@@ -38,6 +38,8 @@ void launch_cconv_fwd(const void*, const void*, const float*, void*, int,
 void launch_cconv_bwd(const void*, const void*, const void*, const float*,
                       void*, void*, float*, float*, int, int, int, int,
                       hipStream_t);
+int launch_gemm_nt(const void*, const void*, void*, int, int, int,
+                   hipStream_t);
 void launch_segsum_exp_fwd(const float*, void*, long long, int, hipStream_t);
 void launch_segsum_exp_bwd(const void*, const float*, float*, long long, int,
                            hipStream_t);
@@ -338,6 +340,18 @@ Tensor segsum_exp_bwd(Tensor g, Tensor cs) {
 }
 
 
+Tensor gemm_nt(Tensor A, Tensor B) {
+  CHECK_BF16_CONTIG(A);
+  CHECK_BF16_CONTIG(B);
+  const int M = A.size(0), K = A.size(1), N = B.size(1);
+  TORCH_CHECK(B.size(0) == K);
+  auto C = torch::empty({M, N}, A.options());
+  const int rc = launch_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr(), M,
+                                N, K, cur_stream());
+  TORCH_CHECK(rc == 0, "gemm_nt: shape not supported (M%256/N%256/K%64)");
+  return C;
+}
+
 // ---- fused SSD scan pieces (see ops/hip/ssd.hip) ----
 // strided 2-D bf16 slice (rows, cols) with unit inner stride
 static long long slice_stride(const Tensor& t, const char* nm) {
@@ -493,6 +507,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_fwd_bwd", &ce_fwd_bwd);
   mod.def("adamw", &adamw);
   mod.def("sq_norm_accum", &sq_norm_accum);
+  mod.def("gemm_nt", &gemm_nt);
   mod.def("ssd_prep_fwd", &ssd_prep_fwd);
   mod.def("ssd_prep_bwd", &ssd_prep_bwd);
   mod.def("ssd_xdt_fwd", &ssd_xdt_fwd);

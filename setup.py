@@ -20,7 +20,8 @@ HIP_DIR = os.path.join(REPO, "fms_fsdp_amd", "ops", "hip")
 BUILD_DIR = os.path.join(REPO, "build", "hip_objs")
 
 HIP_SOURCES = ["rmsnorm.hip", "rope.hip", "swiglu.hip", "cross_entropy.hip",
-               "adamw.hip", "attention.hip", "causal_conv1d.hip", "ssd.hip"]
+               "adamw.hip", "attention.hip", "causal_conv1d.hip", "ssd.hip",
+               "gemm_nt.hip"]
 
 
 def compile_hip_objects():

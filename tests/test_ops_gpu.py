@@ -565,3 +565,15 @@ def test_ssd_fused_scan_gqa_heads():
     o_t, gu_t = run(True)
     assert relerr(o_f, o_t) < 3e-2
     assert relerr(gu_f, gu_t) < 6e-2
+
+
+def test_gemm_nt_exact():
+    """Hand-written dgrad-layout GEMM (documented experiment): must be
+    numerically exact vs hipBLASLt (same fp32 accumulation order class)."""
+    from fms_fsdp_amd import _C
+    torch.manual_seed(0)
+    a = torch.randn(512, 256, device=dev(), dtype=torch.bfloat16) * 0.1
+    b = torch.randn(256, 512, device=dev(), dtype=torch.bfloat16) * 0.1
+    got = _C.gemm_nt(a, b)
+    ref = torch.mm(a, b)
+    assert relerr(got, ref) < 1e-2

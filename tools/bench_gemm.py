@@ -63,5 +63,32 @@ def main():
     print(f"CE chunk x4: {t*4*1e3:.1f} ms/step")
 
 
+def bench_custom():
+    """A/B the hand-written NT-class dgrad kernel vs hipBLASLt."""
+    from fms_fsdp_amd import _C
+    torch.manual_seed(0)
+    print("\n-- custom gemm_nt vs torch.mm (NT dgrad class) --")
+    for (name, m, k, n) in [("proj dgrad", M, 4096, 4096),
+                            ("qkv dgrad", M, 12288, 4096),
+                            ("wg1 dgrad", M, 22016, 4096),
+                            ("w2 dgrad", M, 4096, 11008),
+                            ("ce dx", M, 32000, 4096)]:
+        a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16) * 0.1
+        b = torch.randn(k, n, device="cuda", dtype=torch.bfloat16) * 0.1
+        ref = torch.mm(a, b)
+        got = _C.gemm_nt(a, b)
+        err = ((got.float() - ref.float()).abs().max()
+               / ref.float().abs().max().clamp(min=1e-6)).item()
+        t_lib = bench(lambda: torch.mm(a, b))
+        t_cus = bench(lambda: _C.gemm_nt(a, b))
+        fl = 2 * m * k * n
+        print(f"{name:12s} M{m} K{k:6d} N{n:6d}  lib {t_lib*1e3:7.3f} ms "
+              f"({fl/t_lib/1e12:6.0f} TF)  custom {t_cus*1e3:7.3f} ms "
+              f"({fl/t_cus/1e12:6.0f} TF)  relerr {err:.3e}")
+
+
 if __name__ == "__main__":
-    main()
+    if os.environ.get("GEMM_CUSTOM") == "1":
+        bench_custom()
+    else:
+        main()
