@@ -227,21 +227,26 @@ class Mamba2Mixer(nn.Module):
         scores = torch.einsum("bcqgn,bckgn->bcgqk", Cm, Bm).contiguous()
         sL = ops.ssd_scores_decay(dacs, scores.view(b * nc * g, Q, Q), h, g)
         y_diag = torch.bmm(sL, xdt.view(N, Q, p))          # (N, Q, p) bf16
+        # the whole inter-chunk state chain runs bf16 (W entries are
+        # decays in [0,1], states are sums of bf16-rounded products with
+        # fp32 GEMM accumulation) — halves the traffic of the biggest
+        # remaining torch-side tensors. Only dacs' own chain stays fp32.
         states = torch.einsum(
             "bckgn,bcgrkp->bcgrnp", Bm,
-            xdtd.view(b, nc, g, rep, Q, p)).float().reshape(b, nc, h, n, p)
+            xdtd.view(b, nc, g, rep, Q, p)).reshape(b, nc, h, n, p)
         # inter-chunk recurrence in closed form (see ssd_chunked). The
         # one-chunk shift lives on the TINY (b,h,nc,nc) decay matrix
         # (zero first row) instead of a 167 MB cat on the states tensor.
         G = dacs.view(b, nc, h, Q)[..., -1].permute(0, 2, 1)
         W = torch.exp(segsum(G))
-        Wsh = torch.cat([torch.zeros_like(W[:, :, :1]), W[:, :, :-1]], dim=2)
+        Wsh = torch.cat([torch.zeros_like(W[:, :, :1]), W[:, :, :-1]],
+                        dim=2).to(torch.bfloat16)
         prev = torch.einsum("bhzc,bchnp->bzhnp", Wsh, states)
         # output ordering "bcgqrp" = the bmm-natural layout; the ygate
         # kernel reads it in place
         y_off = torch.einsum(
             "bcqgn,bcgrnp->bcgqrp", Cm,
-            prev.view(b, nc, g, rep, n, p).to(torch.bfloat16))
+            prev.view(b, nc, g, rep, n, p))
         out = ops.ssd_ygate(
             y_diag.reshape(b * l, h * p),
             y_off.contiguous().view(b * l, h * p),
