@@ -450,30 +450,25 @@ __global__ void attn_bwd_delta_kernel(const short* __restrict__ dog,
 }
 
 // ---------------------------------------------------------------------
-// Backward dq: grid over q-tiles; recompute P and dP once, then
-// dq^T = K^T @ dS^T. No atomics: each wave owns its 32 q rows.
-//   P^T  = exp(scale*K@Q^T - lse)    (A = K rows from the k image)
-//   dP^T = V @ dO^T                  (A = V rows from the v image)
-//   dS^T = P^T * (dP^T - delta) * scale
-//   dq^T[dk][q] = sum_key K^T[dk][key] dS^T[key][q]
-//     (A = K^T via tr_read on the SAME k image, B = pack(dS^T))
+// Backward dq, consuming the dS workspace the dkv kernel published:
+//   dq^T[dk][q] = sum_key K^T[dk][key] * dS^T[key][q]
+// No S/dP recompute, no softmax, no lse/delta reads — the backward
+// drops from 7 gemm chains to 5 (dkv computes dS once for both sides).
+// A = K^T via tr reads on the K image; B = dS^T via tr reads on the
+// staged dS tile. grid over q-tiles; each wave owns its 32 q rows.
 // ---------------------------------------------------------------------
 template <int D>
 __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
-    const short* __restrict__ dog, const short* __restrict__ qg,
-    const short* __restrict__ kg, const short* __restrict__ vg,
-    const float* __restrict__ lseg, const float* __restrict__ deltag,
-    short* __restrict__ dqg, int B, int S, int H, int KVH, float scale,
-    long long vstride) {
+    const short* __restrict__ kg, const short* __restrict__ dsw,
+    short* __restrict__ dqg, int B, int S, int H, int KVH) {
   constexpr int KVB = 32;
-  constexpr int NC = D / 16;
   constexpr int NT = D / 32;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // double-buffered subtiled K/V images: stage tile t+1 during tile t's
-  // compute, ONE barrier per tile.
-  constexpr int IMGB = KVB * D * 2;
-#define KIMG(buf) (smem + ((buf) ? 2 * IMGB : 0))
-#define VIMG(buf) (smem + IMGB + ((buf) ? 2 * IMGB : 0))
+  // double-buffered subtiled images: K (KVB x D) + dS (KVB x 128 q)
+  constexpr int KIMGB = KVB * D * 2;
+  constexpr int SIMGB = KVB * 128 * 2;
+#define KIMG(buf) (smem + ((buf) ? KIMGB + SIMGB : 0))
+#define SIMG(buf) (smem + KIMGB + ((buf) ? KIMGB + SIMGB : 0))
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -490,29 +485,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 
   const long long qrow_stride = (long long)H * D;
   const long long krow_stride = (long long)KVH * D;
-  const short* qbase = qg + ((long long)b * S * H + (long long)h) * D;
-  const short* dobase = dog + ((long long)b * S * H + (long long)h) * D;
   const short* kbase = kg + ((long long)b * S * KVH + (long long)kvh) * D;
-  const short* vbase = vg + (long long)b * S * vstride + (long long)kvh * D;
-
-  const float scale2 = scale * 1.4426950408889634f;
-  bf16x8v qb[NC], dob[NC];
-  {
-    const short* qp = qbase + (long long)my_q * qrow_stride + hb * 8;
-    const short* dp = dobase + (long long)my_q * qrow_stride + hb * 8;
-#pragma unroll
-    for (int c = 0; c < NC; ++c) {
-      bf16x8v qv = *(const bf16x8v*)(qp + c * 16);
-#pragma unroll
-      for (int e = 0; e < 8; ++e)   // pre-scale (see fwd kernel)
-        qv[e] = (__bf16)((float)qv[e] * scale2);
-      qb[c] = qv;
-      dob[c] = *(const bf16x8v*)(dp + c * 16);
-    }
-  }
-  const float my_lse2 =
-      lseg[((long long)bh) * S + my_q] * 1.4426950408889634f;
-  const float my_delta = deltag[((long long)bh) * S + my_q];
+  const short* dsbase = dsw + (long long)bh * S * S + q0;
 
   f32x16 accDQ[NT];
 #pragma unroll
@@ -520,76 +494,58 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 
   const int ntiles = (q0 + 128 + KVB - 1) / KVB;
   const int t256 = threadIdx.x;
-  // stage K/V into subtiled images: thread t owns key t&31 and ADJACENT
-  // column chunks (one cacheline per image per thread — a generic
-  // strided-chunk loop here cost dq +16%/dkv +40%, profiles/).
-  // Incremental global pointers + precomputed LDS offsets (see fwd).
+  // staging: K rows (t&31, D cols split over 8 groups) and dS rows
+  // (t&31 keys x 128 q cols split over 8 groups of 16)
   const int skey = t256 & 31;
-  const int sd0 = (t256 >> 5) * ((D == 128) ? 16 : 8);
-  const short* kp_s = kbase + (long long)skey * krow_stride + sd0;
-  const short* vp_s = vbase + (long long)skey * vstride + sd0;
+  const int sgrp = t256 >> 5;            // 0..7
+  const int skd0 = sgrp * (D / 8);
+  const short* kp_s = kbase + (long long)skey * krow_stride + skd0;
   const long long kstep = (long long)KVB * krow_stride;
-  const long long vstep = (long long)KVB * vstride;
-  const int soff0 = SUBT_OFF(skey, sd0, 8);
-  const int soff1 = SUBT_OFF(skey, sd0 + 8, 8);
+  const short* sp_s = dsbase + (long long)skey * S + sgrp * 16;
+  const long long sstep = (long long)KVB * S;
+  const int koff = SUBT_OFF(skey, skd0, 8);
+  const int koff1 = SUBT_OFF(skey, skd0 + 8, 8);
+  const int soff0 = SUBT_OFF(skey, sgrp * 16, 8);
+  const int soff1 = SUBT_OFF(skey, sgrp * 16 + 8, 8);
   auto stage = [&](int buf) {
     if constexpr (D == 128) {
-      *(f32x4*)(KIMG(buf) + soff0) = *(const f32x4*)(kp_s);
-      *(f32x4*)(KIMG(buf) + soff1) = *(const f32x4*)(kp_s + 8);
-      *(f32x4*)(VIMG(buf) + soff0) = *(const f32x4*)(vp_s);
-      *(f32x4*)(VIMG(buf) + soff1) = *(const f32x4*)(vp_s + 8);
-    } else {  // D=64: 32 keys x 8 chunks, exactly one 16B chunk/thread
-      *(f32x4*)(KIMG(buf) + soff0) = *(const f32x4*)(kp_s);
-      *(f32x4*)(VIMG(buf) + soff0) = *(const f32x4*)(vp_s);
+      *(f32x4*)(KIMG(buf) + koff) = *(const f32x4*)(kp_s);
+      *(f32x4*)(KIMG(buf) + koff1) = *(const f32x4*)(kp_s + 8);
+    } else {
+      *(f32x4*)(KIMG(buf) + koff) = *(const f32x4*)(kp_s);
     }
+    *(f32x4*)(SIMG(buf) + soff0) = *(const f32x4*)(sp_s);
+    *(f32x4*)(SIMG(buf) + soff1) = *(const f32x4*)(sp_s + 8);
     kp_s += kstep;
-    vp_s += vstep;
+    sp_s += sstep;
   };
   stage(0);
   __syncthreads();
-  // tr-read bases into the K images (R4 = KVB/4 = 8)
-  const unsigned dq_swz = tr_base<1024>(lane, col, hb);
-  const unsigned kib[2] = {(unsigned)(unsigned long long)KIMG(0) + dq_swz,
-                           (unsigned)(unsigned long long)KIMG(1) + dq_swz};
+  // tr-read bases (R4 = 8 for both images)
+  const unsigned ktr = tr_base<1024>(lane, col, hb);
+  const unsigned kib[2] = {(unsigned)(unsigned long long)KIMG(0) + ktr,
+                           (unsigned)(unsigned long long)KIMG(1) + ktr};
+  // dS image cols = the wave's q window: base col = wid*32
+  const unsigned str = (unsigned)((((lane & 15) * 8) ^ (16 * hb)) +
+                                  hb * 256 + ((col >> 4) << 10) +
+                                  (wid * 32) * 64);
+  const unsigned sib[2] = {(unsigned)(unsigned long long)SIMG(0) + str,
+                           (unsigned)(unsigned long long)SIMG(1) + str};
+
   int cur = 0;
   for (int tile = 0; tile < ntiles; ++tile) {
     if (tile + 1 < ntiles) stage(cur ^ 1);
-    const int kv0 = tile * KVB;
-
-    // S^T and dP^T (A rows = keys, plain subtiled reads)
-    f32x16 accS = (f32x16)(0.f), accDP = (f32x16)(0.f);
-#pragma unroll
-    for (int c = 0; c < NC; ++c) {
-      const int off = SUBT_OFF(col, c * 16 + hb * 8, 8);
-      const bf16x8v ka = *(const bf16x8v*)(KIMG(cur) + off);
-      const bf16x8v va = *(const bf16x8v*)(VIMG(cur) + off);
-      accS = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qb[c], accS, 0, 0, 0);
-      accDP = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dob[c], accDP, 0, 0, 0);
-    }
-
-    float ds[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int key = kv0 + DROW(r, hb);
-      const float pval = exp2f(accS[r] - my_lse2);  // q pre-scaled
-      float v = pval * (accDP[r] - my_delta) * scale;
-      if (key > my_q) v = 0.f;
-      ds[r] = v;
-    }
-
-    // dq^T += K^T @ dS^T : A = tr_read on k image (col dk across key
-    // rows); pipelined immediate-offset reads (see fwd PV loop)
+    // B fragments: dS^T, two 16-key chunks
+    U2x64 bs[2];
+    tr_issue4<0, 128, 512, 512 + 128>(sib[cur], bs[0], bs[1]);
+    tr_wait<0>(bs[0], bs[1]);
     const unsigned kbase_t = kib[cur];
     U2x64 fr[2][2];
     tr_issue4<0, 128, 512, 640>(kbase_t, fr[0][0], fr[0][1]);
-    bf16x8v dsb[2];
-    dsb[0] = pack_pT_chunk(ds);
-    dsb[1] = pack_pT_chunk(ds + 8);
 #pragma unroll
     for (int t = 0; t < NT; ++t) {
       const int pp = t & 1;
       if (t + 1 < NT) {
-        // offsets: (t+1)*2*8*128 = (t+1)*2048 + rg*128
         if (t + 1 == 1)
           tr_issue4<2048, 2048 + 128, 2048 + 512, 2048 + 640>(
               kbase_t, fr[1][0], fr[1][1]);
@@ -603,15 +559,18 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       } else {
         tr_wait<0>(fr[pp][0], fr[pp][1]);
       }
-      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][0].v, dsb[0],
-                                                         accDQ[t], 0, 0, 0);
-      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][1].v, dsb[1],
-                                                         accDQ[t], 0, 0, 0);
+      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][0].v,
+                                                         bs[0].v, accDQ[t],
+                                                         0, 0, 0);
+      accDQ[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fr[pp][1].v,
+                                                         bs[1].v, accDQ[t],
+                                                         0, 0, 0);
     }
     __syncthreads();
     cur ^= 1;
   }
 
+  // dq was accumulated UNSCALED dS? No — dkv stored ds already *scale.
   short* dqp = dqg + ((long long)b * S * H + (long long)h) * D +
                (long long)my_q * qrow_stride;
 #pragma unroll
@@ -627,7 +586,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   }
 }
 #undef KIMG
-#undef VIMG
+#undef SIMG
 
 // ---------------------------------------------------------------------
 // Backward dk/dv: grid over KV tiles; each wave owns 32 keys and walks
@@ -643,7 +602,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const short* __restrict__ dog, const short* __restrict__ qg,
     const short* __restrict__ kg, const short* __restrict__ vg,
     const float* __restrict__ lseg, const float* __restrict__ deltag,
-    void* __restrict__ dkg, void* __restrict__ dvg, int B, int S, int H,
+    void* __restrict__ dkg, void* __restrict__ dvg,
+    short* __restrict__ dsw, int B, int S, int H,
     int KVH, float scale, long long vstride, long long dvstride) {
   constexpr int KVB = 32;   // keys per wave; block = 4 waves = 128 keys
   constexpr int NC = D / 16;
@@ -809,6 +769,21 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       const int key = DROW(r, hb);
       *(short*)(my_p + key * 80 + col * 2) = f2bf(ds[r]);
     }
+    // publish this (32-key x 32-q) dS tile to the global workspace: the
+    // dq kernel consumes it instead of recomputing the S and dP chains
+    // (7 gemm chains across the backward -> 5). Vectorized readback from
+    // the my_p transpose buffer: lane l covers key l&31, 16B q-chunk
+    // (l>>5)*2 + c.
+    {
+      const int key = lane & 31;
+      const long long grow = ((long long)bh * S + kv0 + key) * S + q0;
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        const int qc = ((lane >> 5) * 2 + c) * 8;
+        *(f32x4*)(dsw + grow + qc) =
+            *(const f32x4*)(my_p + key * 80 + qc * 2);
+      }
+    }
     {
       bf16x8v da[2];
 #pragma unroll
@@ -875,7 +850,8 @@ void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
 
 void launch_attn_bwd(const void* do_, const void* q, const void* k,
                      const void* v, const void* o, const float* lse, void* dq,
-                     void* dk, void* dv, float* delta_ws, int B, int S, int H,
+                     void* dk, void* dv, float* delta_ws, void* ds_ws,
+                     int B, int S, int H,
                      int KVH, int D, float scale, int out_bf16,
                      long long vstride, long long dvstride,
                      hipStream_t stream) {
@@ -883,38 +859,37 @@ void launch_attn_bwd(const void* do_, const void* q, const void* k,
   attn_bwd_delta_kernel<<<(int)((rows * 64 + 255) / 256), 256, 0, stream>>>(
       (const short*)do_, (const short*)o, delta_ws, D, S, H, rows);
   dim3 grid(S / 128, B * H);
+  // dkv FIRST: it publishes the dS workspace the dq kernel consumes
   if (D == 128) {
-    const int lds_dq = 4 * 32 * 128 * 2;   // dbuf subtiled k+v images
-    attn_bwd_dq_kernel<128><<<grid, 256, lds_dq, stream>>>(
-        (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
-        lse, delta_ws, (short*)dq, B, S, H, KVH, scale, vstride);
     const int lds_dkv = 4 * 32 * 128 * 2 + 4 * 32 * 128 * 2 + 4 * 32 * 80;
     if (out_bf16)
       attn_bwd_dkv_kernel<128, true><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
-          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale,
-          vstride, dvstride);
+          (const short*)v, lse, delta_ws, dk, dv, (short*)ds_ws, B, S, H,
+          KVH, scale, vstride, dvstride);
     else
       attn_bwd_dkv_kernel<128, false><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
-          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale,
-          vstride, dvstride);
+          (const short*)v, lse, delta_ws, dk, dv, (short*)ds_ws, B, S, H,
+          KVH, scale, vstride, dvstride);
+    const int lds_dq = 2 * (32 * 128 * 2 + 32 * 128 * 2);
+    attn_bwd_dq_kernel<128><<<grid, 256, lds_dq, stream>>>(
+        (const short*)k, (const short*)ds_ws, (short*)dq, B, S, H, KVH);
   } else {
-    const int lds_dq = 4 * 32 * 64 * 2;
-    attn_bwd_dq_kernel<64><<<grid, 256, lds_dq, stream>>>(
-        (const short*)do_, (const short*)q, (const short*)k, (const short*)v,
-        lse, delta_ws, (short*)dq, B, S, H, KVH, scale, vstride);
     const int lds_dkv = 4 * 32 * 64 * 2 + 4 * 32 * 64 * 2 + 4 * 32 * 80;
     if (out_bf16)
       attn_bwd_dkv_kernel<64, true><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
-          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale,
-          vstride, dvstride);
+          (const short*)v, lse, delta_ws, dk, dv, (short*)ds_ws, B, S, H,
+          KVH, scale, vstride, dvstride);
     else
       attn_bwd_dkv_kernel<64, false><<<grid, 256, lds_dkv, stream>>>(
           (const short*)do_, (const short*)q, (const short*)k,
-          (const short*)v, lse, delta_ws, dk, dv, B, S, H, KVH, scale,
-          vstride, dvstride);
+          (const short*)v, lse, delta_ws, dk, dv, (short*)ds_ws, B, S, H,
+          KVH, scale, vstride, dvstride);
+    const int lds_dq = 2 * (32 * 64 * 2 + 32 * 128 * 2);
+    attn_bwd_dq_kernel<64><<<grid, 256, lds_dq, stream>>>(
+        (const short*)k, (const short*)ds_ws, (short*)dq, B, S, H, KVH);
   }
 }
 

@@ -31,8 +31,8 @@ void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, float, long long, hipStream_t);
 void launch_attn_bwd(const void*, const void*, const void*, const void*,
                      const void*, const float*, void*, void*, void*, float*,
-                     int, int, int, int, int, float, int, long long,
-                     long long, hipStream_t);
+                     void*, int, int, int, int, int, float, int,
+                     long long, long long, hipStream_t);
 void launch_cconv_fwd(const void*, const void*, const float*, void*, int,
                       int, int, int, hipStream_t);
 void launch_cconv_bwd(const void*, const void*, const void*, const float*,
@@ -276,10 +276,16 @@ std::tuple<Tensor, Tensor, Tensor> attn_bwd(Tensor do_, Tensor q, Tensor k,
                   : torch::zeros({b, s, kvh, d}, opts);
   }
   auto delta = torch::empty({b, h, s}, q.options().dtype(torch::kFloat32));
+  // dS workspace (b, h, s, s) bf16: written by dkv, consumed by dq —
+  // the backward's P/dP recompute runs once instead of twice. ~2.1 GB
+  // at the 7B shape; the caching allocator reuses it across layers.
+  auto ds_ws = torch::empty({(long long)b * h, (long long)s, (long long)s},
+                            q.options());
   const float scale = 1.f / sqrtf((float)d);
   launch_attn_bwd(do_.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                   o.data_ptr(), lse.data_ptr<float>(), dq.data_ptr(),
-                  dk.data_ptr(), dv.data_ptr(), delta.data_ptr<float>(), b, s,
+                  dk.data_ptr(), dv.data_ptr(), delta.data_ptr<float>(),
+                  ds_ws.data_ptr(), b, s,
                   h, kvh, d, scale, out_bf16 ? 1 : 0, vs, dvs, cur_stream());
   if (out_bf16) return {dq, dk, dv};
   return {dq, dk.to(torch::kBFloat16), dv.to(torch::kBFloat16)};
