@@ -577,3 +577,17 @@ def test_gemm_nt_exact():
     got = _C.gemm_nt(a, b)
     ref = torch.mm(a, b)
     assert relerr(got, ref) < 1e-2
+
+
+def test_speculator_gpu_smoke(tmp_path):
+    """Speculator trainer end-to-end on GPU: stage-1 AND stage-2 (the
+    KV-cached generate path) on the HIP kernels."""
+    from speculator import train_speculator as ts
+    ts.main(model_variant="llama2_125m", use_dummy_dataset=True,
+            batch_size=1, seq_length=256, num_steps=2, report_interval=1,
+            checkpoint_interval=100, n_speculator_heads=2,
+            speculator_width=64, stage2_start_step=1, stage2_batch_size=4,
+            stage2_prompt_length=16, stage2_seq_length=32,
+            model_path="/nonexistent", ckpt_save_path=str(tmp_path),
+            ckpt_load_path=str(tmp_path), vocab_size=256,
+            learning_rate=1e-4, sharding_strategy="fsdp")
