@@ -46,3 +46,29 @@ def test_param_counts():
         m = Llama(get_model_config("llama2_70b"))
     n = m.param_count()
     assert 6.6e10 < n < 7.2e10, n
+
+
+def test_all_registry_variants_construct_on_meta():
+    """Every llama variant in the registry builds on the meta device with
+    a sane parameter count (catches config typos without materializing
+    70B anywhere)."""
+    import torch
+    from fms_fsdp_amd.config import get_model_config
+    from fms_fsdp_amd.models import Llama
+    expected = {"llama2_1.4b": 1.25e9, "llama2_7b": 6.5e9,
+                "llama2_13b": 12e9, "llama2_34b": 32e9, "llama2_70b": 65e9,
+                "llama3_194m": 0.15e9, "llama3_1.8b": 1.5e9,
+                "llama3_3.2b": 3.0e9, "llama3_8b": 7e9, "llama3_70b": 65e9}
+    for variant, lo in expected.items():
+        for suffix in ("", "_4k"):
+            name = variant + suffix
+            try:
+                cfg = get_model_config(name)
+            except (KeyError, ValueError):
+                if suffix:
+                    continue    # _4k variants only exist where the
+                raise           # reference defines them
+            with torch.device("meta"):
+                m = Llama(cfg)
+            n = sum(p.numel() for p in m.parameters())
+            assert n >= lo, (name, n)
