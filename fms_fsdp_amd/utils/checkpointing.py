@@ -193,13 +193,35 @@ class Checkpointer:
         # num_workers=0 if exact resume through Checkpointer alone matters.
         payloads = []
         if is_writer:
-            model_sd = {u.name: u.master_shard.cpu() for u in model.all_units}
+            if hasattr(model, "_pool_master_shard"):
+                # pooled shard storage: THREE large contiguous D2H copies
+                # instead of 3 x n_units small ones (the small-copy path
+                # measured multi-second training stalls per checkpoint at
+                # 7B). The per-unit views are cloned in the writer thread
+                # (torch.save would otherwise serialize the whole pool
+                # into every file).
+                mcpu = model._pool_master_shard.cpu()
+                eacpu = model._pool_exp_avg.cpu()
+                evcpu = model._pool_exp_avg_sq.cpu()
+                sl, off = {}, 0
+                for u in model.all_units:
+                    sl[u.name] = slice(off, off + u.shard_size)
+                    off += u.shard_size
+                model_sd = {u.name: mcpu[sl[u.name]] for u in model.all_units}
+                opt_units = {u.name: {"exp_avg": eacpu[sl[u.name]],
+                                      "exp_avg_sq": evcpu[sl[u.name]]}
+                             for u in model.all_units}
+            else:
+                model_sd = {u.name: u.master_shard.cpu()
+                            for u in model.all_units}
+                opt_units = {u.name: {"exp_avg": u.exp_avg.cpu(),
+                                      "exp_avg_sq": u.exp_avg_sq.cpu()}
+                             for u in model.all_units}
             payloads.append((model_sd,
                              os.path.join(out, f"model_{shard_rank}_of_{S}.pth")))
-            opt_sd = {"step": optimizer.step_count, "lr": optimizer.param_groups[0]["lr"],
-                      "units": {u.name: {"exp_avg": u.exp_avg.cpu(),
-                                         "exp_avg_sq": u.exp_avg_sq.cpu()}
-                                for u in model.all_units}}
+            opt_sd = {"step": optimizer.step_count,
+                      "lr": optimizer.param_groups[0]["lr"],
+                      "units": opt_units}
             payloads.append((opt_sd,
                              os.path.join(out, f"optim_{shard_rank}_of_{S}.pth")))
         if dataloader is not None and hasattr(dataloader.dataset, "state_dict"):
@@ -229,9 +251,21 @@ class Checkpointer:
                 dist.barrier()
         return out
 
+    @staticmethod
+    def _clone_tree(obj):
+        """Deep-clone tensor leaves so torch.save serializes only each
+        view's data, not the pooled backing storage."""
+        if torch.is_tensor(obj):
+            return obj.clone() if obj._base is not None or \
+                obj.numel() != obj.untyped_storage().nbytes() // obj.element_size() \
+                else obj
+        if isinstance(obj, dict):
+            return {k: Checkpointer._clone_tree(v) for k, v in obj.items()}
+        return obj
+
     def _write_payloads(self, payloads, meta, out, t0):
         for obj, path in payloads:
-            torch.save(obj, path)
+            torch.save(self._clone_tree(obj), path)
             base = os.path.basename(path)
             if base.startswith(("model_", "optim_")):
                 with open(path + ".done", "w"):
