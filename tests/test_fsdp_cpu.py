@@ -620,3 +620,59 @@ def test_world8_save_reshard2_resume(tmp_path):
     status, loss3 = q2.get()
     assert status == "ok", loss3
     assert abs(loss3 - ref3) < 1e-5, (loss3, ref3)
+
+
+def _reshard3_worker(rank, world, port, tmpdir, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from fms_fsdp_amd.models import Llama, LlamaBlock
+        from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+        from fms_fsdp_amd.utils.checkpointing import Checkpointer
+        torch.manual_seed(0)
+        m = Llama(_small_cfg())
+        m.reset_parameters()
+        sm = ShardedModel(m, LlamaBlock, sharding_strategy="fsdp",
+                          param_dtype=torch.float32)
+        opt = ShardedAdamW(sm, lr=1e-3)
+        ck = Checkpointer(tmpdir, 3, "fsdp", rank, rank)
+        _, _, _, step, _, resuming = ck.load(sm, opt, None, path="")
+        assert step == 2 and resuming
+        g = torch.Generator().manual_seed(42)
+        x = torch.randint(0, 128, (2, 32), generator=g)
+        y = torch.randint(0, 128, (2, 32), generator=g)
+        opt.zero_grad()
+        loss = sm(x, labels=y)
+        loss.backward()
+        sm.clip_grad_norm_(1.0)
+        opt.step()
+        if rank == 0:
+            q.put(("ok", loss.item()))
+    except Exception as e:
+        q.put(("err", f"{type(e).__name__}: {e}"))
+        raise
+    finally:
+        dist.destroy_process_group()
+
+
+def test_save2_resume3_odd_world(tmp_path):
+    """NON-DIVISOR resharding: save at world=2, resume at world=3 (the
+    per-unit flat padding differs between shard worlds — offsets are
+    S-independent so the reshard must still be exact)."""
+    ref3 = _single_process_reference()[2]
+    tmpdir = str(tmp_path)
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    # _ckpt8_worker trains the SAME model/recipe as the single-process
+    # reference (3 layers, clipped) before saving
+    mp.spawn(_ckpt8_worker, args=(2, PORT + 45, tmpdir, q), nprocs=2,
+             join=True)
+    status, _ = q.get()
+    assert status == "ok", _
+    q2 = ctx.SimpleQueue()
+    mp.spawn(_reshard3_worker, args=(3, PORT + 46, tmpdir, q2), nprocs=3,
+             join=True)
+    status, loss3 = q2.get()
+    assert status == "ok", loss3
+    assert abs(loss3 - ref3) < 1e-5, (loss3, ref3)
