@@ -173,3 +173,43 @@ def test_policy_world2_matches_single(policy):
     assert status == "ok", got
     for a, b in zip(ref, got):
         assert abs(a - b) < 2e-3, (ref, got)
+
+
+def test_fp16_overflow_skip_and_recover():
+    """An injected overflow step must be SKIPPED (params/moments
+    untouched, scale backed off) and training must continue normally
+    afterwards — the full scaler path through the sharded runtime."""
+    sm, opt, tc = _mk("fp16", seed=3)
+    scaler = DynamicGradScaler(enabled=True, init_scale=8.0)
+    g = torch.Generator().manual_seed(5)
+    x = torch.randint(0, 128, (2, 32), generator=g)
+    y = torch.randint(0, 128, (2, 32), generator=g)
+
+    # one normal step
+    opt.zero_grad()
+    scaler.scale_loss(sm(x, labels=y)).backward()
+    _, stepped = scaler.clip_and_step(sm, opt, 1.0)
+    assert stepped
+    before = [u.master_shard.clone() for u in sm.all_units]
+    moments = [u.exp_avg.clone() for u in sm.all_units]
+    step_count = opt.step_count
+
+    # inject an overflow into the grads
+    opt.zero_grad()
+    scaler.scale_loss(sm(x, labels=y)).backward()
+    sm.all_units[0].grad_shard.view(-1)[0] = float("inf")
+    gnorm, stepped = scaler.clip_and_step(sm, opt, 1.0)
+    assert not stepped and not bool(torch.isfinite(gnorm))
+    assert scaler.scale == 4.0                       # backed off
+    for u, b, m in zip(sm.all_units, before, moments):
+        assert torch.equal(u.master_shard, b), u.name # update skipped
+        assert torch.equal(u.exp_avg, m), u.name      # moments untouched
+    assert opt.step_count == step_count               # bias corr. frozen
+
+    # recovery: next clean step applies normally
+    opt.zero_grad()
+    loss = sm(x, labels=y)
+    scaler.scale_loss(loss).backward()
+    _, stepped = scaler.clip_and_step(sm, opt, 1.0)
+    assert stepped and torch.isfinite(loss)
+    assert not torch.equal(sm.all_units[0].master_shard, before[0])
