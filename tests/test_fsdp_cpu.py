@@ -676,3 +676,63 @@ def test_save2_resume3_odd_world(tmp_path):
     status, loss3 = q2.get()
     assert status == "ok", loss3
     assert abs(loss3 - ref3) < 1e-5, (loss3, ref3)
+
+
+def _hsdp_ckpt_worker(rank, world, port, tmpdir, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LOCAL_WORLD_SIZE"] = "2"
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from fms_fsdp_amd.models import Llama, LlamaBlock
+        from fms_fsdp_amd.parallel import ShardedModel, ShardedAdamW
+        from fms_fsdp_amd.utils.checkpointing import Checkpointer
+
+        class FakeLoader:
+            class DS:
+                def state_dict(self):
+                    return {"pos": 7}
+            dataset = DS()
+
+        torch.manual_seed(0)
+        m = Llama(_small_cfg())
+        m.reset_parameters()
+        sm = ShardedModel(m, LlamaBlock, sharding_strategy="hsdp",
+                          param_dtype=torch.float32, intra_node_size=2)
+        opt = ShardedAdamW(sm, lr=1e-3)
+        g = torch.Generator().manual_seed(42)
+        x = torch.randint(0, 128, (2, 32), generator=g)
+        y = torch.randint(0, 128, (2, 32), generator=g)
+        opt.zero_grad()
+        sm(x, labels=y).backward()
+        opt.step()
+        Checkpointer(tmpdir, 3, "hsdp", rank, rank).save(
+            1, sm, opt, FakeLoader())
+        if rank == 0:
+            q.put(("ok", None))
+    except Exception as e:
+        q.put(("err", f"{type(e).__name__}: {e}"))
+        raise
+    finally:
+        dist.destroy_process_group()
+
+
+def test_hsdp_checkpoint_write_dedup(tmp_path):
+    """HSDP at 2 'nodes' x 2 'GPUs': only the replicate-rank-0 shard
+    group writes model/optim shards (reference checkpointing_utils.py
+    :137-141), while EVERY rank writes its loader state."""
+    tmpdir = str(tmp_path)
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_hsdp_ckpt_worker, args=(4, PORT + 47, tmpdir, q), nprocs=4,
+             join=True)
+    status, _ = q.get()
+    assert status == "ok", _
+    files = sorted(os.listdir(os.path.join(tmpdir, "checkpoints",
+                                           "step_1_ckp")))
+    models = [f for f in files if f.startswith("model_") and
+              f.endswith(".pth")]
+    loaders = [f for f in files if f.startswith("loader_state_")]
+    assert models == ["model_0_of_2.pth", "model_1_of_2.pth"], files
+    assert len(loaders) == 4, files
+    assert "metadata.pth" in files
