@@ -517,8 +517,8 @@ def _ref70():
 
 
 @pytest.mark.parametrize("strat,intra,port", [
-    ("fsdp", None, PORT + 41),
-    ("hsdp", 4, PORT + 42),   # 2 'nodes' x 4 'GPUs'
+    ("fsdp", None, PORT + 61),
+    ("hsdp", 4, PORT + 62),   # 2 'nodes' x 4 'GPUs'
 ])
 def test_world8_70b_execution_path(strat, intra, port):
     """World-8 gloo: meta-streamed init + reshard_after_forward +
@@ -610,12 +610,12 @@ def test_world8_save_reshard2_resume(tmp_path):
     tmpdir = str(tmp_path)
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    mp.spawn(_ckpt8_worker, args=(8, PORT + 43, tmpdir, q), nprocs=8,
+    mp.spawn(_ckpt8_worker, args=(8, PORT + 63, tmpdir, q), nprocs=8,
              join=True)
     status, _ = q.get()
     assert status == "ok", _
     q2 = ctx.SimpleQueue()
-    mp.spawn(_resume2_worker, args=(2, PORT + 44, tmpdir, q2), nprocs=2,
+    mp.spawn(_resume2_worker, args=(2, PORT + 64, tmpdir, q2), nprocs=2,
              join=True)
     status, loss3 = q2.get()
     assert status == "ok", loss3
@@ -666,12 +666,12 @@ def test_save2_resume3_odd_world(tmp_path):
     q = ctx.SimpleQueue()
     # _ckpt8_worker trains the SAME model/recipe as the single-process
     # reference (3 layers, clipped) before saving
-    mp.spawn(_ckpt8_worker, args=(2, PORT + 45, tmpdir, q), nprocs=2,
+    mp.spawn(_ckpt8_worker, args=(2, PORT + 65, tmpdir, q), nprocs=2,
              join=True)
     status, _ = q.get()
     assert status == "ok", _
     q2 = ctx.SimpleQueue()
-    mp.spawn(_reshard3_worker, args=(3, PORT + 46, tmpdir, q2), nprocs=3,
+    mp.spawn(_reshard3_worker, args=(3, PORT + 66, tmpdir, q2), nprocs=3,
              join=True)
     status, loss3 = q2.get()
     assert status == "ok", loss3
@@ -724,7 +724,7 @@ def test_hsdp_checkpoint_write_dedup(tmp_path):
     tmpdir = str(tmp_path)
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    mp.spawn(_hsdp_ckpt_worker, args=(4, PORT + 47, tmpdir, q), nprocs=4,
+    mp.spawn(_hsdp_ckpt_worker, args=(4, PORT + 67, tmpdir, q), nprocs=4,
              join=True)
     status, _ = q.get()
     assert status == "ok", _
